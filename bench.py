@@ -1,0 +1,205 @@
+#!/usr/bin/env python3
+"""Benchmark harness for the MI355X-native IMHN pose framework.
+
+Measures the flagship 4-stage IMHN @512x512 bf16 on synthetic data with
+random-init weights (no network access in this environment):
+
+  * --mode train (default): full training step — synthetic batch -> forward ->
+    focal-L2 loss -> backward (with overlapped RCCL all-reduce when N > 1) ->
+    fused SGD step. Metric: aggregate train_images_per_sec over all ranks
+    (weak scaling: fixed per-GPU batch).
+  * --mode infer: forward-only inference, batch 4, last-stack output — the
+    reference's headline configuration (38.5 FPS on a 2080 Ti,
+    BASELINE.md / reference test_inference_speed.py). Metric: FPS.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+         --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints exactly one JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from improved_body_parts_amd.config import GetConfig, TrainingOpt  # noqa: E402
+from improved_body_parts_amd.data import SyntheticPoseDataset  # noqa: E402
+from improved_body_parts_amd.engine import FusedSGD  # noqa: E402
+from improved_body_parts_amd.models import Network, NetworkEval  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--mode", choices=["train", "infer"], default="train")
+    p.add_argument("--batch", type=int, default=None,
+                   help="per-GPU batch (default: 16 train / 4 infer)")
+    p.add_argument("--nstack", type=int, default=4)
+    p.add_argument("--input", type=int, default=512)
+    p.add_argument("--no-bf16", action="store_true")
+    p.add_argument("--allow-eager", action="store_true",
+                   help="permit eager fallback if the HIP extension is absent")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    if args.allow_eager:
+        os.environ["IBP_AMD_ALLOW_EAGER"] = "1"
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world_size > 1
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    if distributed:
+        import torch.distributed as dist
+        dist.init_process_group(backend="nccl" if use_cuda else "gloo",
+                                init_method="env://")
+
+    batch = args.batch or (16 if args.mode == "train" else 4)
+    bf16 = use_cuda and not args.no_bf16
+    dtype = torch.bfloat16 if bf16 else torch.float32
+
+    cfg_name = {512: "Canonical", 384: "Canonical384", 768: "Canonical768"}.get(
+        args.input, "Canonical")
+    config = GetConfig(cfg_name)
+    opt = TrainingOpt(nstack=args.nstack, batch_size=batch,
+                      nstack_weight=[1] * args.nstack)
+
+    torch.manual_seed(1234)  # same random init on all ranks
+
+    if args.mode == "train":
+        model = Network(opt, config, bn=True, dist=True).to(device)
+        if bf16:
+            model = model.bfloat16()
+            for m in model.modules():
+                if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+                    m.float()
+        model.train()
+        optimizer = FusedSGD(model.parameters(), lr=opt.learning_rate * world_size,
+                             momentum=opt.momentum, weight_decay=opt.weight_decay)
+        reducer = None
+        if distributed:
+            from improved_body_parts_amd.parallel import GradReducer
+            reducer = GradReducer(model, broadcast_parameters=True)
+
+        # pre-stage a few synthetic batches on device (different per rank)
+        ds = SyntheticPoseDataset(config, length=world_size * 2, seed=17)
+        batches = []
+        for b in range(2):
+            idx = rank * 2 + b
+            img, mm, hm = ds[idx]
+            img = img[None].expand(batch, -1, -1, -1).contiguous()
+            mm = mm[None].expand(batch, -1, -1, -1).contiguous()
+            hm = hm[None].expand(batch, -1, -1, -1).contiguous()
+            batches.append(tuple(t.to(device=device, dtype=dtype) for t in (img, mm, hm)))
+
+        def step(i):
+            b = batches[i % len(batches)]
+            if reducer is not None:
+                reducer.zero_grad()
+            else:
+                optimizer.zero_grad(set_to_none=False)
+            loss = model(b)
+            loss.backward()
+            if reducer is not None:
+                reducer.finalize()
+            optimizer.step()
+            return loss
+
+        metric_name = "train_images_per_sec"
+        vs_baseline = None
+        higher = True
+    else:
+        model = NetworkEval(opt, config, bn=True).to(device)
+        if bf16:
+            model = model.bfloat16()
+            for m in model.modules():
+                if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+                    m.float()
+        model.eval()
+        img = torch.rand(batch, config.height, config.width, 3,
+                         device=device, dtype=dtype)
+
+        @torch.no_grad()
+        def step(i):
+            out = model(img)
+            return out[-1][0]  # last stack, scale 0 (reference evaluate.py:126)
+
+        metric_name = "fps_512_infer"
+        vs_baseline = None  # filled below from the 38.5 FPS headline
+        higher = True
+
+    def sync():
+        if distributed:
+            import torch.distributed as dist
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        step(i)
+    sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    if distributed:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], device=device if use_cuda else None)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t)
+
+    images = batch * args.steps * world_size
+    value = images / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+    if args.mode == "infer":
+        vs_baseline = value / 38.5  # reference headline: 38.5 FPS @512^2 batch 4
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": metric_name,
+            "value": round(value, 2),
+            "unit": "images/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": higher,
+            "scaling": "weak",
+            "vs_baseline": round(vs_baseline, 3) if vs_baseline is not None else None,
+            "dtype": "bf16" if bf16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"{args.nstack}-stage IMHN @{args.input}x{args.input}",
+                "mode": args.mode,
+                "global_batch": batch * world_size,
+                "input": args.input,
+                "parallelism": f"dp{world_size}",
+            },
+        }), flush=True)
+
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

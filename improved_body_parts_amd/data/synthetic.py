@@ -1,0 +1,108 @@
+"""Synthetic pose dataset: random images + procedurally generated skeletons -> GT.
+
+There is no network access in the build/bench environment, so the benchmark and
+the training smoke paths run on synthetic data of the exact shape the reference
+trains on: image (H, W, 3) in [0,1], mask_miss (1, H/4, W/4),
+heatmaps (num_layers, H/4, W/4). Skeletons are sampled as plausible 2D stick
+figures so the GT generator exercises every keypoint/limb channel.
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+from torch.utils.data import Dataset
+
+from .heatmapper import Heatmapper
+
+# canonical-order template skeleton in a ~[0,1]x[0,1] body frame
+# (part order: see config.canonical._PARTS)
+_TEMPLATE = np.array([
+    [0.50, 0.10],  # nose
+    [0.50, 0.22],  # neck
+    [0.38, 0.24],  # Rsho
+    [0.33, 0.40],  # Relb
+    [0.30, 0.55],  # Rwri
+    [0.62, 0.24],  # Lsho
+    [0.67, 0.40],  # Lelb
+    [0.70, 0.55],  # Lwri
+    [0.42, 0.55],  # Rhip
+    [0.41, 0.75],  # Rkne
+    [0.40, 0.95],  # Rank
+    [0.58, 0.55],  # Lhip
+    [0.59, 0.75],  # Lkne
+    [0.60, 0.95],  # Lank
+    [0.46, 0.07],  # Reye
+    [0.54, 0.07],  # Leye
+    [0.42, 0.10],  # Rear
+    [0.58, 0.10],  # Lear
+], dtype=np.float32)
+
+
+def sample_people(rng: np.random.Generator, width: int, height: int,
+                  max_people: int = 4) -> np.ndarray:
+    """Return (P, 18, 3) canonical joints with visibility flags."""
+    n = int(rng.integers(1, max_people + 1))
+    people = []
+    for _ in range(n):
+        scale = rng.uniform(0.25, 0.9) * height
+        cx = rng.uniform(0.15, 0.85) * width
+        cy = rng.uniform(0.25, 0.75) * height
+        jitter = rng.normal(0, 0.02, _TEMPLATE.shape).astype(np.float32)
+        pts = (_TEMPLATE - [0.5, 0.5] + jitter) * scale
+        ang = rng.uniform(-0.4, 0.4)
+        rot = np.array([[np.cos(ang), -np.sin(ang)], [np.sin(ang), np.cos(ang)]],
+                       np.float32)
+        pts = pts @ rot.T + [cx, cy]
+        vis = np.ones((18, 1), np.float32)
+        # randomly drop some annotations (visibility 2 = not marked)
+        drop = rng.random(18) < 0.15
+        vis[drop] = 2
+        people.append(np.concatenate([pts, vis], axis=1))
+    return np.stack(people, axis=0)
+
+
+class SyntheticPoseDataset(Dataset):
+    """Deterministic (per-index) synthetic samples matching the training contract
+    of reference data/mydataset.py: __getitem__ -> (image (H,W,3), mask_miss
+    (1,h,w), heatmaps (C,h,w)) as float32 torch tensors."""
+
+    def __init__(self, config, length: int = 1024, seed: int = 0, max_people: int = 4):
+        self.config = config
+        self.length = length
+        self.seed = seed
+        self.max_people = max_people
+        self.heatmapper = Heatmapper(config)
+
+    def __len__(self):
+        return self.length
+
+    def generate(self, index: int):
+        cfg = self.config
+        rng = np.random.default_rng(self.seed * 1_000_003 + index)
+        img = rng.random((cfg.height, cfg.width, 3), dtype=np.float32)
+        joints = sample_people(rng, cfg.width, cfg.height, self.max_people)
+        h, w = cfg.mask_shape
+        mask_all = np.zeros((h, w), np.float32)
+        # person boxes become the mask_all foreground
+        for p in joints:
+            marked = p[:, 2] < 2
+            if not marked.any():
+                continue
+            xs, ys = p[marked, 0] / cfg.stride, p[marked, 1] / cfg.stride
+            x0, x1 = int(max(xs.min() - 2, 0)), int(min(xs.max() + 2, w))
+            y0, y1 = int(max(ys.min() - 2, 0)), int(min(ys.max() + 2, h))
+            mask_all[y0:y1, x0:x1] = 1.0
+        # random unannotated region -> mask_miss zero patch
+        mask_miss = np.ones((h, w), np.float32)
+        if rng.random() < 0.5 and w >= 12 and h >= 12:
+            mw = int(rng.integers(2, max(w // 4, 3)))
+            mh = int(rng.integers(2, max(h // 4, 3)))
+            mx, my = int(rng.integers(0, w - mw)), int(rng.integers(0, h - mh))
+            mask_miss[my:my + mh, mx:mx + mw] = 0.0
+        heatmaps = self.heatmapper.create_heatmaps(joints, mask_all)
+        return img, mask_miss[None], heatmaps, joints
+
+    def __getitem__(self, index: int):
+        img, mask_miss, heatmaps, _ = self.generate(index)
+        return (torch.from_numpy(img), torch.from_numpy(mask_miss),
+                torch.from_numpy(np.ascontiguousarray(heatmaps)))

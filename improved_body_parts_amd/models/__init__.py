@@ -1,0 +1,15 @@
+"""Model layer: IMHN stacked hourglass + loss.
+
+Public API matches the reference's models/ package (posenet.py, layers_transposed.py,
+loss_model.py) so user code and checkpoints carry over unchanged.
+"""
+from .layers import (Conv, Residual, BasicResidual, DilatedConv, Backbone,
+                     Hourglass, SELayer)
+from .posenet import PoseNet, Network, NetworkEval, Merge, Features
+from .loss import MultiTaskLoss, MultiTaskLossParallel
+
+__all__ = [
+    "Conv", "Residual", "BasicResidual", "DilatedConv", "Backbone", "Hourglass",
+    "SELayer", "PoseNet", "Network", "NetworkEval", "Merge", "Features",
+    "MultiTaskLoss", "MultiTaskLossParallel",
+]
