@@ -20,8 +20,11 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 
 def _sources():
-    return sorted(glob.glob(os.path.join(CSRC, "*.hip"))) + \
-        sorted(glob.glob(os.path.join(CSRC, "*.cpp")))
+    # torch's hipify pass writes shadow copies named *_hip.hip next to the
+    # originals; exclude them or a rebuild would compile every file twice.
+    hips = [s for s in sorted(glob.glob(os.path.join(CSRC, "*.hip")))
+            if not s.endswith("_hip.hip")]
+    return hips + sorted(glob.glob(os.path.join(CSRC, "*.cpp")))
 
 
 def _needs_rebuild(sources):

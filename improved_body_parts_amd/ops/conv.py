@@ -1,0 +1,160 @@
+"""Fused conv + BatchNorm + LeakyReLU (+ residual) on the HIP path.
+
+The convolution itself goes through :mod:`.conv_kernels` (hand-written MFMA
+implicit-GEMM for the supported shapes); BN statistics, the scale/shift+act
+epilogue and the whole BN backward run as the fused CDNA4 kernels of
+bn_act.hip. Tensors stay NHWC (torch channels_last) end to end.
+
+Replaces: nn.Conv2d -> nn.BatchNorm2d -> nn.LeakyReLU chains
+(reference models/layers_transposed.py:90-120) and their autograd.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from ._backend import hip_extension
+
+LEAKY_SLOPE = 0.01
+_CL = torch.channels_last
+
+
+def _to_cl(t):
+    return t.contiguous(memory_format=_CL)
+
+
+def _conv_forward(x, weight, bias, stride, padding, dilation):
+    """Convolution forward dispatch: MFMA implicit-GEMM kernel when the shape is
+    supported, library fallback otherwise (first iterations / odd shapes)."""
+    from . import conv_kernels
+    y = conv_kernels.conv_fwd(x, weight, stride, padding, dilation)
+    if y is None:
+        y = F.conv2d(x, weight, None, stride, padding, dilation)
+        y = _to_cl(y)
+    if bias is not None:
+        y = y + bias.view(1, -1, 1, 1)
+    return y
+
+
+def _conv_dgrad(dy, weight, x_shape, stride, padding, dilation):
+    from . import conv_kernels
+    dx = conv_kernels.conv_dgrad(dy, weight, x_shape, stride, padding, dilation)
+    if dx is None:
+        dx = torch.nn.grad.conv2d_input(x_shape, weight, dy, stride, padding,
+                                        dilation)
+        dx = _to_cl(dx)
+    return dx
+
+
+def _conv_wgrad(x, dy, w_shape, stride, padding, dilation):
+    from . import conv_kernels
+    dw = conv_kernels.conv_wgrad(x, dy, w_shape, stride, padding, dilation)
+    if dw is None:
+        dw = torch.nn.grad.conv2d_weight(x, w_shape, dy, stride, padding,
+                                         dilation)
+    return dw
+
+
+class ConvBnActFn(torch.autograd.Function):
+    """y = leaky( bn( conv(x, w) ) (+ residual) ), all fused on device."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, gamma, beta, residual,
+                stride, padding, dilation, act, training, bn_mod):
+        ext = hip_extension()
+        x = _to_cl(x)
+        y_conv = _conv_forward(x, weight, bias if gamma is None else None,
+                               stride, padding, dilation)
+        C = y_conv.shape[1]
+
+        if gamma is not None:
+            if training:
+                sums, sumsq = ext.bn_stats(y_conv, C)
+                M = y_conv.numel() // C
+                mean = sums / M
+                var = (sumsq / M - mean * mean).clamp_(min=0.0)
+                with torch.no_grad():
+                    mom = bn_mod.momentum if bn_mod.momentum is not None else 0.1
+                    unbiased = var * (M / max(M - 1, 1))
+                    bn_mod.running_mean.mul_(1 - mom).add_(mean, alpha=mom)
+                    bn_mod.running_var.mul_(1 - mom).add_(unbiased, alpha=mom)
+                    bn_mod.num_batches_tracked += 1
+            else:
+                mean = bn_mod.running_mean.float()
+                var = bn_mod.running_var.float()
+            invstd = torch.rsqrt(var + bn_mod.eps)
+            scale = (gamma.float() * invstd)
+            shift = (beta.float() - mean * scale)
+        elif act or residual is not None:
+            mean = invstd = None
+            scale = torch.ones(C, device=x.device, dtype=torch.float32)
+            shift = torch.zeros(C, device=x.device, dtype=torch.float32)
+        else:
+            # plain conv (the 1x1 heads): no epilogue pass needed
+            mean = invstd = scale = shift = None
+
+        if scale is not None:
+            res_cl = _to_cl(residual) if residual is not None else None
+            y = ext.bn_act_fwd(y_conv, scale, shift, res_cl, LEAKY_SLOPE, act)
+        else:
+            y = y_conv
+
+        ctx.save_for_backward(x, weight, gamma, y_conv, y, mean, invstd)
+        ctx.conf = (stride, padding, dilation, act, training,
+                    residual is not None, bias is not None and gamma is None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_extension()
+        x, weight, gamma, y_conv, y, mean, invstd = ctx.saved_tensors
+        stride, padding, dilation, act, training, has_res, has_bias = ctx.conf
+        C = y_conv.shape[1]
+        dy = _to_cl(dy)
+        has_bn = gamma is not None
+
+        if not has_bn and not act and not has_res and not has_bias:
+            # plain conv fast path (the 1x1 heads without bias)
+            dx = _conv_dgrad(dy, weight, x.shape, stride, padding, dilation) \
+                if ctx.needs_input_grad[0] else None
+            dw = _conv_wgrad(x, dy, weight.shape, stride, padding, dilation) \
+                if ctx.needs_input_grad[1] else None
+            return (dx, dw, None, None, None, None,
+                    None, None, None, None, None, None)
+
+        dpre, sum_dpre, sum_dxhat = ext.bn_act_bwd(
+            dy, y, y_conv, mean if has_bn else None, invstd if has_bn else None,
+            LEAKY_SLOPE, act, has_bn, C)
+
+        dres = dpre if has_res else None
+        if has_bn:
+            if training:
+                dconv = ext.bn_act_bwd_apply(dpre, y_conv, mean, invstd,
+                                             gamma.float(), sum_dpre, sum_dxhat, C)
+            else:
+                dconv = ext.bn_act_bwd_apply(dpre, y_conv, mean, invstd,
+                                             gamma.float(), None, None, C)
+            dgamma = sum_dxhat.to(gamma.dtype)
+            dbeta = sum_dpre.to(gamma.dtype)
+        else:
+            dconv = dpre
+            dgamma = dbeta = None
+
+        dbias = sum_dpre.to(weight.dtype) if has_bias else None
+        dx = _conv_dgrad(dconv, weight, x.shape, stride, padding, dilation) \
+            if ctx.needs_input_grad[0] else None
+        dw = _conv_wgrad(x, dconv, weight.shape, stride, padding, dilation) \
+            if ctx.needs_input_grad[1] else None
+        return (dx, dw, dbias, dgamma, dbeta, dres,
+                None, None, None, None, None, None)
+
+
+def conv_bn_act_hip(x, conv, bn, act: bool, residual=None, training: bool = False):
+    """Module-level entry used by models.layers: pulls parameters out of the
+    nn.Conv2d / nn.BatchNorm2d containers and runs the fused function."""
+    gamma = bn.weight if bn is not None else None
+    beta = bn.bias if bn is not None else None
+    return ConvBnActFn.apply(
+        x, conv.weight, conv.bias, gamma, beta, residual,
+        conv.stride, conv.padding, conv.dilation, act,
+        training and (bn is not None and bn.training), bn)

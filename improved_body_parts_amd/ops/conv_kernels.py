@@ -1,0 +1,128 @@
+"""MFMA implicit-GEMM convolution dispatch.
+
+The CDNA4 kernels (csrc/conv_mfma.hip) implement NHWC bf16 convolution as an
+implicit GEMM on the bf16 matrix cores:
+    C[M=N*Ho*Wo][N=Cout] = A[M][K=KH*KW*Cin] @ B[K][Cout]
+with fwd, dgrad (stride-1: conv with 180-rotated transposed weights) and
+split-K wgrad. Weights are re-packed to [KH*KW*Cin][Cout] (Cout contiguous)
+per forward; the pack is cached against the weight tensor's version counter.
+
+Each entry point returns ``None`` when the shape is outside the kernel's
+envelope, and the caller falls back to the library conv.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+
+from ._backend import hip_extension
+
+_CL = torch.channels_last
+
+# shapes supported by the MFMA path: stride-1 any dilation, and stride-2 fwd
+_DISABLE = os.environ.get("IBP_AMD_DISABLE_MFMA_CONV") == "1"
+
+
+def _supported(x, weight, stride, padding, dilation, for_grad=False):
+    if _DISABLE:
+        return False
+    if x.dtype != torch.bfloat16:
+        return False
+    kh, kw = weight.shape[2], weight.shape[3]
+    if padding[0] != (kh - 1) // 2 * dilation[0] or \
+       padding[1] != (kw - 1) // 2 * dilation[1]:
+        return False
+    if for_grad and (stride[0] != 1 or stride[1] != 1):
+        return False
+    if stride[0] not in (1, 2) or stride[0] != stride[1]:
+        return False
+    return True
+
+
+_pack_cache = {}
+
+
+def packed_weight(weight: torch.Tensor) -> torch.Tensor:
+    """[Cout, Cin, kh, kw] -> bf16 [kh*kw*Cin, Cout] contiguous, cached."""
+    key = id(weight)
+    entry = _pack_cache.get(key)
+    ver = weight._version
+    if entry is not None and entry[0] == ver:
+        return entry[1]
+    w = weight.detach().to(torch.bfloat16)
+    packed = w.permute(2, 3, 1, 0).reshape(-1, weight.shape[0]).contiguous()
+    _pack_cache[key] = (ver, packed)
+    return packed
+
+
+def packed_weight_dgrad(weight: torch.Tensor) -> torch.Tensor:
+    """Weights for dgrad-as-conv: rotate 180° spatially and swap Cin/Cout ->
+    [kh*kw*Cout, Cin]."""
+    key = (id(weight), "dgrad")
+    entry = _pack_cache.get(key)
+    ver = weight._version
+    if entry is not None and entry[0] == ver:
+        return entry[1]
+    w = weight.detach().to(torch.bfloat16)
+    w = torch.flip(w, dims=(2, 3))          # rotate the kernel
+    packed = w.permute(2, 3, 0, 1).reshape(-1, weight.shape[1]).contiguous()
+    _pack_cache[key] = (ver, packed)
+    return packed
+
+
+def conv_fwd(x, weight, stride, padding, dilation):
+    if not _supported(x, weight, stride, padding, dilation):
+        return None
+    ext = hip_extension()
+    if not hasattr(ext, "conv_mfma_fwd"):
+        return None
+    x = x.contiguous(memory_format=_CL)
+    n, cin, h, w_ = x.shape
+    cout, _, kh, kw = weight.shape
+    ho = (h + 2 * padding[0] - dilation[0] * (kh - 1) - 1) // stride[0] + 1
+    wo = (w_ + 2 * padding[1] - dilation[1] * (kw - 1) - 1) // stride[1] + 1
+    y = ext.conv_mfma_fwd(x, packed_weight(weight), n, h, w_, cin, cout,
+                          kh, kw, stride[0], padding[0], padding[1],
+                          dilation[0], dilation[1], ho, wo)
+    return y.permute(0, 3, 1, 2)  # NHWC buffer -> NCHW view (channels_last)
+
+
+def conv_dgrad(dy, weight, x_shape, stride, padding, dilation):
+    if not _supported(dy, weight, stride, padding, dilation, for_grad=True):
+        return None
+    ext = hip_extension()
+    if not hasattr(ext, "conv_mfma_fwd"):
+        return None
+    dy = dy.contiguous(memory_format=_CL)
+    n, cout, h, w_ = dy.shape
+    cin = x_shape[1]
+    kh, kw = weight.shape[2], weight.shape[3]
+    # stride-1 'same' conv: dx = conv(dy, rot180(W)^T), same padding/dilation
+    dx = ext.conv_mfma_fwd(dy, packed_weight_dgrad(weight), n, h, w_, cout, cin,
+                           kh, kw, 1, padding[0], padding[1],
+                           dilation[0], dilation[1], h, w_)
+    return dx.permute(0, 3, 1, 2)
+
+
+def conv_wgrad(x, dy, w_shape, stride, padding, dilation):
+    if _DISABLE or x.dtype != torch.bfloat16:
+        return None
+    ext = hip_extension()
+    if not hasattr(ext, "conv_mfma_wgrad"):
+        return None
+    kh, kw = w_shape[2], w_shape[3]
+    if padding[0] != (kh - 1) // 2 * dilation[0] or \
+       padding[1] != (kw - 1) // 2 * dilation[1]:
+        return None
+    x = x.contiguous(memory_format=_CL)
+    dy = dy.contiguous(memory_format=_CL)
+    n, cin, h, w_ = x.shape
+    cout = dy.shape[1]
+    ho, wo = dy.shape[2], dy.shape[3]
+    dw_kkc = ext.conv_mfma_wgrad(x, dy, n, h, w_, cin, cout, kh, kw,
+                                 stride[0], padding[0], padding[1],
+                                 dilation[0], dilation[1], ho, wo)
+    # [kh*kw*Cin][Cout] fp32 -> [Cout, Cin, kh, kw] in the weight's dtype
+    dw = dw_kkc.reshape(kh, kw, cin, cout).permute(3, 2, 0, 1)
+    return dw.to(x.dtype).contiguous()

@@ -1,0 +1,68 @@
+// Python bindings for the CDNA4 kernel library.
+#include <torch/extension.h>
+
+using torch::Tensor;
+
+// bn_act.hip
+std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C);
+Tensor bn_act_fwd(const Tensor& x, const Tensor& scale, const Tensor& shift,
+                  const c10::optional<Tensor>& residual, double slope, bool act);
+std::vector<Tensor> bn_act_bwd(const Tensor& dy, const Tensor& y, const Tensor& x,
+                               const c10::optional<Tensor>& mean,
+                               const c10::optional<Tensor>& invstd,
+                               double slope, bool act, bool need_xhat, int64_t C);
+Tensor bn_act_bwd_apply(const Tensor& dpre, const Tensor& x, const Tensor& mean,
+                        const Tensor& invstd, const Tensor& gamma,
+                        const c10::optional<Tensor>& sum_dpre,
+                        const c10::optional<Tensor>& sum_dxhat, int64_t C);
+
+// spatial.hip
+std::vector<Tensor> maxpool2x2_fwd(const Tensor& x, int64_t N, int64_t H,
+                                   int64_t W, int64_t C);
+Tensor maxpool2x2_bwd(const Tensor& dy, const Tensor& arg, int64_t N, int64_t H,
+                      int64_t W, int64_t C);
+Tensor upsample2x_fwd(const Tensor& x, int64_t N, int64_t H, int64_t W, int64_t C);
+Tensor upsample2x_bwd(const Tensor& dy, int64_t N, int64_t H, int64_t W, int64_t C);
+Tensor se_reduce(const Tensor& a, const c10::optional<Tensor>& b, int64_t N,
+                 int64_t HW, int64_t C);
+Tensor se_scale(const Tensor& x, const Tensor& s, const c10::optional<Tensor>& addc,
+                int64_t N, int64_t HW, int64_t C);
+
+// loss.hip
+Tensor focal_l2_fwd(const Tensor& pred, const Tensor& gt, const Tensor& mask,
+                    int64_t heat_start, int64_t bkg_start, int64_t gamma,
+                    double mtw, double ktw, double alpha, double beta);
+Tensor focal_l2_bwd(const Tensor& pred, const Tensor& gt, const Tensor& mask,
+                    const Tensor& stack_gscale, int64_t heat_start,
+                    int64_t bkg_start, int64_t gamma, double mtw, double ktw,
+                    double alpha, double beta);
+
+// sgd.hip
+void fused_sgd(const Tensor& chunk_table, double lr, double momentum,
+               double weight_decay, int64_t dtype_tag);
+
+// postproc.hip
+Tensor heatmap_nms(const Tensor& heat, double thre);
+std::vector<Tensor> collect_peaks(const Tensor& nmsed, const Tensor& smoothed,
+                                  int64_t radius, int64_t max_peaks);
+Tensor limb_scores(const Tensor& paf, const Tensor& peaks, const Tensor& cand_idx,
+                   int64_t mid_num, double thre2);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_stats", &bn_stats, "per-channel sum/sumsq of an [M][C] view");
+  m.def("bn_act_fwd", &bn_act_fwd, "fused scale/shift (+res) (+leaky)");
+  m.def("bn_act_bwd", &bn_act_bwd, "dpre + per-channel reductions");
+  m.def("bn_act_bwd_apply", &bn_act_bwd_apply, "BN backward input grad");
+  m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
+  m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
+  m.def("upsample2x_fwd", &upsample2x_fwd);
+  m.def("upsample2x_bwd", &upsample2x_bwd);
+  m.def("se_reduce", &se_reduce);
+  m.def("se_scale", &se_scale);
+  m.def("focal_l2_fwd", &focal_l2_fwd);
+  m.def("focal_l2_bwd", &focal_l2_bwd);
+  m.def("fused_sgd", &fused_sgd);
+  m.def("heatmap_nms", &heatmap_nms);
+  m.def("collect_peaks", &collect_peaks);
+  m.def("limb_scores", &limb_scores);
+}

@@ -1,0 +1,277 @@
+// Fused BatchNorm + LeakyReLU (+ residual add) over NHWC activations.
+//
+// Replaces the reference's nn.BatchNorm2d + nn.LeakyReLU pairs (cuDNN ops,
+// reference models/layers_transposed.py:90-120) with single-pass CDNA4 kernels:
+// the normalisation is folded to y = act(scale[c] * x + shift[c] (+ res)) with
+// fp32 statistics, bf16 activations, 16-byte vector accesses per lane.
+//
+// Exposed ops (see bindings.cpp):
+//   bn_stats        : per-channel sum / sum-of-squares of an [M][C] view
+//   bn_act_fwd      : fused scale/shift + optional residual + optional leaky
+//   bn_act_bwd      : dpre = dy * act'(y); per-channel Σdpre, Σdpre*xhat
+//   bn_act_bwd_apply: dx from dpre + the reduced sums (training BN backward)
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace ibp {
+
+// --------------------------------------------------------------------------
+// per-channel statistics: sums[c] = Σ_m x[m][c], sumsq[c] = Σ_m x[m][c]^2
+// grid.y tiles channels by blockDim.x; grid.x strides rows; fp32 atomics.
+// --------------------------------------------------------------------------
+template <typename T>
+__global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sums,
+                                float* __restrict__ sumsq, long long M, int C) {
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, q = 0.f;
+  for (long long m = blockIdx.x; m < M; m += gridDim.x) {
+    float v = ldf(x + m * C + c);
+    s += v;
+    q += v * v;
+  }
+  atomicAdd(&sums[c], s);
+  atomicAdd(&sumsq[c], q);
+}
+
+// --------------------------------------------------------------------------
+// y = act(scale[c]*x + shift[c] (+ res)); bf16 path moves 8 elems per lane.
+// --------------------------------------------------------------------------
+typedef unsigned short ushort8 __attribute__((ext_vector_type(8)));
+
+__global__ void bn_act_fwd_bf16v8(const unsigned short* __restrict__ x,
+                                  const unsigned short* __restrict__ res,
+                                  unsigned short* __restrict__ y,
+                                  const float* __restrict__ scale,
+                                  const float* __restrict__ shift,
+                                  long long total8, int C8, float slope, int act) {
+  const ushort8* xv = reinterpret_cast<const ushort8*>(x);
+  const ushort8* rv = reinterpret_cast<const ushort8*>(res);
+  ushort8* yv = reinterpret_cast<ushort8*>(y);
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total8;
+       i += (long long)gridDim.x * blockDim.x) {
+    int cbase = (int)((i % C8) * 8);
+    ushort8 xi = xv[i];
+    ushort8 out;
+    if (res != nullptr) {
+      ushort8 ri = rv[i];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = us2f(xi[j]) * scale[cbase + j] + shift[cbase + j] + us2f(ri[j]);
+        out[j] = f2us(act ? leaky(v, slope) : v);
+      }
+    } else {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = us2f(xi[j]) * scale[cbase + j] + shift[cbase + j];
+        out[j] = f2us(act ? leaky(v, slope) : v);
+      }
+    }
+    yv[i] = out;
+  }
+}
+
+template <typename T>
+__global__ void bn_act_fwd_scalar(const T* __restrict__ x, const T* __restrict__ res,
+                                  T* __restrict__ y, const float* __restrict__ scale,
+                                  const float* __restrict__ shift, long long total,
+                                  int C, float slope, int act) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float v = ldf(x + i) * scale[c] + shift[c];
+    if (res != nullptr) v += ldf(res + i);
+    stf(y + i, act ? leaky(v, slope) : v);
+  }
+}
+
+// --------------------------------------------------------------------------
+// backward part 1: dpre = dy * (act ? (y > 0 ? 1 : slope) : 1)
+//                  sum_dpre[c] += dpre ; sum_dxhat[c] += dpre * (x-mean)*invstd
+// --------------------------------------------------------------------------
+template <typename T, bool NEED_XHAT>
+__global__ void bn_act_bwd_reduce_kernel(
+    const T* __restrict__ dy, const T* __restrict__ y, const T* __restrict__ x,
+    T* __restrict__ dpre_out, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ sum_dpre,
+    float* __restrict__ sum_dxhat, long long M, int C, float slope, int act) {
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, sx = 0.f;
+  float mu = NEED_XHAT ? mean[c] : 0.f;
+  float is = NEED_XHAT ? invstd[c] : 0.f;
+  for (long long m = blockIdx.x; m < M; m += gridDim.x) {
+    long long i = m * C + c;
+    float g = ldf(dy + i);
+    if (act) {
+      float yy = ldf(y + i);
+      g = yy > 0.f ? g : g * slope;
+    }
+    stf(dpre_out + i, g);
+    s += g;
+    if (NEED_XHAT) sx += g * (ldf(x + i) - mu) * is;
+  }
+  atomicAdd(&sum_dpre[c], s);
+  if (NEED_XHAT) atomicAdd(&sum_dxhat[c], sx);
+}
+
+// --------------------------------------------------------------------------
+// backward part 2 (training BN):
+//   dx = gamma*invstd * (dpre - sum_dpre/M - xhat * sum_dxhat/M)
+// eval BN: dx = gamma*invstd*dpre  (pass sums = nullptr)
+// --------------------------------------------------------------------------
+template <typename T>
+__global__ void bn_act_bwd_apply_kernel(
+    const T* __restrict__ dpre, const T* __restrict__ x, T* __restrict__ dx,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ sum_dpre,
+    const float* __restrict__ sum_dxhat, long long M, int C) {
+  long long total = M * C;
+  float invM = 1.0f / (float)M;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float g = ldf(dpre + i);
+    float is = invstd[c];
+    float w = gamma[c] * is;
+    if (sum_dpre != nullptr) {
+      float xhat = (ldf(x + i) - mean[c]) * is;
+      g = g - sum_dpre[c] * invM - xhat * sum_dxhat[c] * invM;
+    }
+    stf(dx + i, w * g);
+  }
+}
+
+}  // namespace ibp
+
+// ===========================================================================
+// host wrappers
+// ===========================================================================
+using torch::Tensor;
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
+  TORCH_CHECK(x_mc.is_cuda() && x_mc.is_contiguous());
+  long long M = x_mc.numel() / C;
+  auto opts = x_mc.options().dtype(torch::kFloat32);
+  Tensor sums = torch::zeros({C}, opts);
+  Tensor sumsq = torch::zeros({C}, opts);
+  dim3 block(256);
+  int rows = (int)std::min<long long>((M + 63) / 64, 1024);
+  dim3 grid(rows, (C + 255) / 256);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      x_mc.scalar_type(), "bn_stats", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(ibp::bn_stats_kernel<T>, grid, block, 0, cur_stream(),
+                       reinterpret_cast<const T*>(x_mc.data_ptr()),
+                       sums.data_ptr<float>(), sumsq.data_ptr<float>(), M, (int)C);
+  });
+  return {sums, sumsq};
+}
+
+Tensor bn_act_fwd(const Tensor& x, const Tensor& scale, const Tensor& shift,
+                  const c10::optional<Tensor>& residual, double slope, bool act) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  int C = (int)scale.numel();
+  long long total = x.numel();
+  Tensor y = torch::empty_like(x);
+  const void* resp = nullptr;
+  if (residual.has_value()) {
+    TORCH_CHECK(residual->is_contiguous() && residual->numel() == total);
+    resp = residual->data_ptr();
+  }
+  dim3 block(256);
+  if (x.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0 &&
+      total % 8 == 0) {
+    long long total8 = total / 8;
+    dim3 grid(ibp::grid_1d(total8, 256, 8192));
+    hipLaunchKernelGGL(ibp::bn_act_fwd_bf16v8, grid, block, 0, cur_stream(),
+                       reinterpret_cast<const unsigned short*>(x.data_ptr()),
+                       reinterpret_cast<const unsigned short*>(resp),
+                       reinterpret_cast<unsigned short*>(y.data_ptr()),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       total8, C / 8, (float)slope, act ? 1 : 0);
+  } else {
+    dim3 grid(ibp::grid_1d(total, 256, 8192));
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+        x.scalar_type(), "bn_act_fwd", [&] {
+      using T = scalar_t;
+      hipLaunchKernelGGL(ibp::bn_act_fwd_scalar<T>, grid, block, 0, cur_stream(),
+                         reinterpret_cast<const T*>(x.data_ptr()),
+                         reinterpret_cast<const T*>(resp),
+                         reinterpret_cast<T*>(y.data_ptr()),
+                         scale.data_ptr<float>(), shift.data_ptr<float>(),
+                         total, C, (float)slope, act ? 1 : 0);
+    });
+  }
+  return y;
+}
+
+std::vector<Tensor> bn_act_bwd(const Tensor& dy, const Tensor& y, const Tensor& x,
+                               const c10::optional<Tensor>& mean,
+                               const c10::optional<Tensor>& invstd,
+                               double slope, bool act, bool need_xhat, int64_t C) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && y.is_contiguous());
+  long long M = dy.numel() / C;
+  auto fopts = dy.options().dtype(torch::kFloat32);
+  Tensor dpre = torch::empty_like(dy);
+  Tensor sum_dpre = torch::zeros({C}, fopts);
+  Tensor sum_dxhat = torch::zeros({C}, fopts);
+  dim3 block(256);
+  int rows = (int)std::min<long long>((M + 63) / 64, 1024);
+  dim3 grid(rows, (C + 255) / 256);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      dy.scalar_type(), "bn_act_bwd", [&] {
+    using T = scalar_t;
+    if (need_xhat) {
+      hipLaunchKernelGGL((ibp::bn_act_bwd_reduce_kernel<T, true>), grid, block, 0,
+                         cur_stream(),
+                         reinterpret_cast<const T*>(dy.data_ptr()),
+                         reinterpret_cast<const T*>(y.data_ptr()),
+                         reinterpret_cast<const T*>(x.data_ptr()),
+                         reinterpret_cast<T*>(dpre.data_ptr()),
+                         mean->data_ptr<float>(), invstd->data_ptr<float>(),
+                         sum_dpre.data_ptr<float>(), sum_dxhat.data_ptr<float>(),
+                         M, (int)C, (float)slope, act ? 1 : 0);
+    } else {
+      hipLaunchKernelGGL((ibp::bn_act_bwd_reduce_kernel<T, false>), grid, block, 0,
+                         cur_stream(),
+                         reinterpret_cast<const T*>(dy.data_ptr()),
+                         reinterpret_cast<const T*>(y.data_ptr()),
+                         reinterpret_cast<const T*>(x.data_ptr()),
+                         reinterpret_cast<T*>(dpre.data_ptr()),
+                         nullptr, nullptr,
+                         sum_dpre.data_ptr<float>(), sum_dxhat.data_ptr<float>(),
+                         M, (int)C, (float)slope, act ? 1 : 0);
+    }
+  });
+  return {dpre, sum_dpre, sum_dxhat};
+}
+
+Tensor bn_act_bwd_apply(const Tensor& dpre, const Tensor& x, const Tensor& mean,
+                        const Tensor& invstd, const Tensor& gamma,
+                        const c10::optional<Tensor>& sum_dpre,
+                        const c10::optional<Tensor>& sum_dxhat, int64_t C) {
+  long long M = dpre.numel() / C;
+  Tensor dx = torch::empty_like(dpre);
+  dim3 block(256);
+  dim3 grid(ibp::grid_1d(dpre.numel(), 256, 8192));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      dpre.scalar_type(), "bn_act_bwd_apply", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(ibp::bn_act_bwd_apply_kernel<T>, grid, block, 0, cur_stream(),
+                       reinterpret_cast<const T*>(dpre.data_ptr()),
+                       reinterpret_cast<const T*>(x.data_ptr()),
+                       reinterpret_cast<T*>(dx.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(),
+                       sum_dpre.has_value() ? sum_dpre->data_ptr<float>() : nullptr,
+                       sum_dxhat.has_value() ? sum_dxhat->data_ptr<float>() : nullptr,
+                       M, (int)C);
+  });
+  return dx;
+}

@@ -1,0 +1,69 @@
+// Fused multi-tensor SGD (momentum + weight decay) with fp32 master weights.
+//
+// Replaces the Apex FusedSGD path the reference mentions
+// (train_distributed.py:121-125) and Apex O1's master-weight machinery:
+// for bf16 parameters the optimizer holds fp32 master weights and the kernel
+// updates (master, momentum) in fp32 then stores the bf16 copy — ONE kernel
+// launch for the whole model instead of ~1850 per-tensor ops.
+//
+//   g = grad (+ wd * w) ; m = mu * m + g ; w -= lr * m
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace ibp {
+
+constexpr int CHUNK = 1 << 16;  // elements per block-chunk
+
+struct SgdChunk {
+  void* param;        // model param (bf16 or fp32)
+  void* grad;         // grad, same dtype as param
+  float* momentum;    // fp32
+  float* master;      // fp32 master (nullptr when param is fp32)
+  long long offset;   // element offset of this chunk within the tensor
+  int count;          // elements in this chunk
+};
+
+template <typename T>
+__global__ void fused_sgd_kernel(const SgdChunk* __restrict__ chunks,
+                                 float lr, float mu, float wd) {
+  const SgdChunk ck = chunks[blockIdx.x];
+  T* p = reinterpret_cast<T*>(ck.param) + ck.offset;
+  const T* g = reinterpret_cast<const T*>(ck.grad) + ck.offset;
+  float* m = ck.momentum + ck.offset;
+  float* w = ck.master ? ck.master + ck.offset : nullptr;
+  for (int i = threadIdx.x; i < ck.count; i += blockDim.x) {
+    float wv = w ? w[i] : ldf(p + i);
+    float gv = ldf(g + i) + wd * wv;
+    float mv = mu * m[i] + gv;
+    m[i] = mv;
+    wv -= lr * mv;
+    if (w) w[i] = wv;
+    stf(p + i, wv);
+  }
+}
+
+}  // namespace ibp
+
+using torch::Tensor;
+
+// chunk table packed on host into an int64 tensor [n][6]:
+// (param_ptr, grad_ptr, momentum_ptr, master_ptr, offset, count)
+void fused_sgd(const Tensor& chunk_table, double lr, double momentum,
+               double weight_decay, int64_t dtype_tag) {
+  TORCH_CHECK(chunk_table.is_cuda() && chunk_table.dtype() == torch::kInt64);
+  int n = (int)chunk_table.size(0);
+  static_assert(sizeof(ibp::SgdChunk) == 6 * 8, "chunk layout mismatch");
+  const ibp::SgdChunk* chunks =
+      reinterpret_cast<const ibp::SgdChunk*>(chunk_table.data_ptr<int64_t>());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  dim3 block(256), grid(n);
+  if (dtype_tag == 0) {  // bf16 params
+    hipLaunchKernelGGL(ibp::fused_sgd_kernel<__hip_bfloat16>, grid, block, 0,
+                       stream, chunks, (float)lr, (float)momentum,
+                       (float)weight_decay);
+  } else {  // fp32 params
+    hipLaunchKernelGGL(ibp::fused_sgd_kernel<float>, grid, block, 0, stream,
+                       chunks, (float)lr, (float)momentum, (float)weight_decay);
+  }
+}

@@ -1,0 +1,245 @@
+// NHWC spatial kernels: 2x2 max-pool, x2 nearest upsample, squeeze-excitation.
+//
+// Replace the reference's nn.MaxPool2d(2,2) / nn.Upsample(nearest) / SELayer
+// (reference models/layers_transposed.py:209-210, :285-306). All kernels are
+// memory-bound; bf16 moves 8 channels per lane where the layout allows.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace ibp {
+
+// ---- maxpool 2x2 stride 2 -------------------------------------------------
+// out[n][ho][wo][c] = max over the 2x2 window; argmax stored as 2 bits for bwd.
+template <typename T>
+__global__ void maxpool2x2_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                      unsigned char* __restrict__ arg,
+                                      int N, int H, int W, int C) {
+  int Ho = H >> 1, Wo = W >> 1;
+  long long total = (long long)N * Ho * Wo * C;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long long p = i / C;
+    int wo = (int)(p % Wo);
+    long long q = p / Wo;
+    int ho = (int)(q % Ho);
+    int n = (int)(q / Ho);
+    const T* base = x + (((long long)n * H + 2 * ho) * W + 2 * wo) * C + c;
+    float v00 = ldf(base);
+    float v01 = ldf(base + C);
+    float v10 = ldf(base + (long long)W * C);
+    float v11 = ldf(base + (long long)W * C + C);
+    float m = v00;
+    int a = 0;
+    if (v01 > m) { m = v01; a = 1; }
+    if (v10 > m) { m = v10; a = 2; }
+    if (v11 > m) { m = v11; a = 3; }
+    stf(y + i, m);
+    arg[i] = (unsigned char)a;
+  }
+}
+
+template <typename T>
+__global__ void maxpool2x2_bwd_kernel(const T* __restrict__ dy,
+                                      const unsigned char* __restrict__ arg,
+                                      T* __restrict__ dx,
+                                      int N, int H, int W, int C) {
+  // iterate over INPUT positions: each belongs to exactly one window
+  int Ho = H >> 1, Wo = W >> 1;
+  long long total = (long long)N * H * W * C;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long long p = i / C;
+    int w = (int)(p % W);
+    long long q = p / W;
+    int h = (int)(q % H);
+    int n = (int)(q / H);
+    if (h >= 2 * Ho || w >= 2 * Wo) { stf(dx + i, 0.f); continue; }
+    int ho = h >> 1, wo = w >> 1;
+    int pos = ((h & 1) << 1) | (w & 1);
+    long long o = (((long long)n * Ho + ho) * Wo + wo) * C + c;
+    stf(dx + i, arg[o] == pos ? ldf(dy + o) : 0.f);
+  }
+}
+
+// ---- nearest x2 upsample ---------------------------------------------------
+template <typename T>
+__global__ void upsample2x_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                      int N, int H, int W, int C) {
+  int Ho = H * 2, Wo = W * 2;
+  long long total = (long long)N * Ho * Wo * C;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long long p = i / C;
+    int wo = (int)(p % Wo);
+    long long q = p / Wo;
+    int ho = (int)(q % Ho);
+    int n = (int)(q / Ho);
+    long long src = (((long long)n * H + (ho >> 1)) * W + (wo >> 1)) * C + c;
+    stf(y + i, ldf(x + src));
+  }
+}
+
+template <typename T>
+__global__ void upsample2x_bwd_kernel(const T* __restrict__ dy, T* __restrict__ dx,
+                                      int N, int H, int W, int C) {
+  // dx[h][w] = sum of the 4 dys that sampled it (H, W are INPUT dims)
+  int Ho = H * 2, Wo = W * 2;
+  long long total = (long long)N * H * W * C;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long long p = i / C;
+    int w = (int)(p % W);
+    long long q = p / W;
+    int h = (int)(q % H);
+    int n = (int)(q / H);
+    const T* base = dy + (((long long)n * Ho + 2 * h) * Wo + 2 * w) * C + c;
+    float s = ldf(base) + ldf(base + C) + ldf(base + (long long)Wo * C)
+            + ldf(base + (long long)Wo * C + C);
+    stf(dx + i, s);
+  }
+}
+
+// ---- squeeze-excitation helpers -------------------------------------------
+// channel reduction r[n][c] = sum_hw a[n][h][w][c] * (b ? b[n][h][w][c] : 1)
+template <typename T, bool PROD>
+__global__ void se_reduce_kernel(const T* __restrict__ a, const T* __restrict__ b,
+                                 float* __restrict__ out, int N, long long HW, int C) {
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  int n = blockIdx.z;
+  const T* pa = a + (long long)n * HW * C;
+  const T* pb = PROD ? b + (long long)n * HW * C : nullptr;
+  float s = 0.f;
+  for (long long m = blockIdx.x; m < HW; m += gridDim.x) {
+    float v = ldf(pa + m * C + c);
+    if (PROD) v *= ldf(pb + m * C + c);
+    s += v;
+  }
+  atomicAdd(&out[(long long)n * C + c], s);
+}
+
+// y = x * s[n][c] (+ optional add[n][c] broadcast)
+template <typename T>
+__global__ void se_scale_kernel(const T* __restrict__ x, const float* __restrict__ s,
+                                const float* __restrict__ addc, T* __restrict__ y,
+                                int N, long long HW, int C) {
+  long long total = (long long)N * HW * C;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int n = (int)(i / (HW * C));
+    float v = ldf(x + i) * s[(long long)n * C + c];
+    if (addc != nullptr) v += addc[(long long)n * C + c];
+    stf(y + i, v);
+  }
+}
+
+}  // namespace ibp
+
+using torch::Tensor;
+static inline hipStream_t cur_stream2() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+std::vector<Tensor> maxpool2x2_fwd(const Tensor& x, int64_t N, int64_t H, int64_t W,
+                                   int64_t C) {
+  Tensor y = torch::empty({N, H / 2, W / 2, C}, x.options());
+  Tensor arg = torch::empty({N, H / 2, W / 2, C},
+                            x.options().dtype(torch::kUInt8));
+  dim3 block(256), grid(ibp::grid_1d(y.numel(), 256, 8192));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      x.scalar_type(), "maxpool2x2_fwd", [&] {
+    hipLaunchKernelGGL(ibp::maxpool2x2_fwd_kernel<scalar_t>, grid, block, 0,
+                       cur_stream2(), reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                       reinterpret_cast<scalar_t*>(y.data_ptr()),
+                       arg.data_ptr<unsigned char>(), (int)N, (int)H, (int)W, (int)C);
+  });
+  return {y, arg};
+}
+
+Tensor maxpool2x2_bwd(const Tensor& dy, const Tensor& arg, int64_t N, int64_t H,
+                      int64_t W, int64_t C) {
+  Tensor dx = torch::empty({N, H, W, C}, dy.options());
+  dim3 block(256), grid(ibp::grid_1d(dx.numel(), 256, 8192));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      dy.scalar_type(), "maxpool2x2_bwd", [&] {
+    hipLaunchKernelGGL(ibp::maxpool2x2_bwd_kernel<scalar_t>, grid, block, 0,
+                       cur_stream2(), reinterpret_cast<const scalar_t*>(dy.data_ptr()),
+                       arg.data_ptr<unsigned char>(),
+                       reinterpret_cast<scalar_t*>(dx.data_ptr()),
+                       (int)N, (int)H, (int)W, (int)C);
+  });
+  return dx;
+}
+
+Tensor upsample2x_fwd(const Tensor& x, int64_t N, int64_t H, int64_t W, int64_t C) {
+  Tensor y = torch::empty({N, H * 2, W * 2, C}, x.options());
+  dim3 block(256), grid(ibp::grid_1d(y.numel(), 256, 8192));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      x.scalar_type(), "upsample2x_fwd", [&] {
+    hipLaunchKernelGGL(ibp::upsample2x_fwd_kernel<scalar_t>, grid, block, 0,
+                       cur_stream2(), reinterpret_cast<const scalar_t*>(x.data_ptr()),
+                       reinterpret_cast<scalar_t*>(y.data_ptr()),
+                       (int)N, (int)H, (int)W, (int)C);
+  });
+  return y;
+}
+
+Tensor upsample2x_bwd(const Tensor& dy, int64_t N, int64_t H, int64_t W, int64_t C) {
+  Tensor dx = torch::empty({N, H, W, C}, dy.options());
+  dim3 block(256), grid(ibp::grid_1d(dx.numel(), 256, 8192));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      dy.scalar_type(), "upsample2x_bwd", [&] {
+    hipLaunchKernelGGL(ibp::upsample2x_bwd_kernel<scalar_t>, grid, block, 0,
+                       cur_stream2(), reinterpret_cast<const scalar_t*>(dy.data_ptr()),
+                       reinterpret_cast<scalar_t*>(dx.data_ptr()),
+                       (int)N, (int)H, (int)W, (int)C);
+  });
+  return dx;
+}
+
+// r[n][c] = sum_hw a (*b); used for SE global-avg-pool (b absent) and
+// SE backward ds (b = the other operand)
+Tensor se_reduce(const Tensor& a, const c10::optional<Tensor>& b, int64_t N,
+                 int64_t HW, int64_t C) {
+  Tensor out = torch::zeros({N, C}, a.options().dtype(torch::kFloat32));
+  dim3 block(256);
+  int rows = (int)std::min<long long>((HW + 63) / 64, 512);
+  dim3 grid(rows, (C + 255) / 256, N);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      a.scalar_type(), "se_reduce", [&] {
+    using T = scalar_t;
+    if (b.has_value()) {
+      hipLaunchKernelGGL((ibp::se_reduce_kernel<T, true>), grid, block, 0,
+                         cur_stream2(), reinterpret_cast<const T*>(a.data_ptr()),
+                         reinterpret_cast<const T*>(b->data_ptr()),
+                         out.data_ptr<float>(), (int)N, HW, (int)C);
+    } else {
+      hipLaunchKernelGGL((ibp::se_reduce_kernel<T, false>), grid, block, 0,
+                         cur_stream2(), reinterpret_cast<const T*>(a.data_ptr()),
+                         nullptr, out.data_ptr<float>(), (int)N, HW, (int)C);
+    }
+  });
+  return out;
+}
+
+Tensor se_scale(const Tensor& x, const Tensor& s, const c10::optional<Tensor>& addc,
+                int64_t N, int64_t HW, int64_t C) {
+  Tensor y = torch::empty_like(x);
+  dim3 block(256), grid(ibp::grid_1d(x.numel(), 256, 8192));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      x.scalar_type(), "se_scale", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(ibp::se_scale_kernel<T>, grid, block, 0, cur_stream2(),
+                       reinterpret_cast<const T*>(x.data_ptr()),
+                       s.data_ptr<float>(),
+                       addc.has_value() ? addc->data_ptr<float>() : nullptr,
+                       reinterpret_cast<T*>(y.data_ptr()), (int)N, HW, (int)C);
+  });
+  return y;
+}

@@ -1,0 +1,315 @@
+"""GPU numerics tests: every HIP kernel vs the plain-PyTorch fp32 oracle.
+
+Run on an MI355X: python -m pytest tests -m gpu -x -q
+"""
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from improved_body_parts_amd import ops
+    from improved_body_parts_amd.ops import _backend
+
+CL = torch.channels_last
+
+
+def _assert_close(got, ref, rtol, atol, what=""):
+    got = got.detach().float().cpu()
+    ref = ref.detach().float().cpu()
+    err = (got - ref).abs().max().item()
+    denom = ref.abs().max().item() + 1e-8
+    assert torch.allclose(got, ref, rtol=rtol, atol=atol), \
+        f"{what}: max abs err {err:.3e} (ref max {denom:.3e})"
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _require_ext():
+    assert _backend.hip_available(), \
+        "HIP extension must be built and loaded on a GPU box"
+
+
+# ---------------------------------------------------------------------------
+# fused conv+bn+act vs eager fp32
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype,tol", [(torch.float32, 1e-4), (torch.bfloat16, 5e-2)])
+@pytest.mark.parametrize("training", [True, False])
+def test_conv_bn_act_forward_backward(dtype, tol, training):
+    torch.manual_seed(0)
+    n, cin, cout, hw = 2, 16, 32, 16
+    conv = torch.nn.Conv2d(cin, cout, 3, padding=1, bias=False)
+    bn = torch.nn.BatchNorm2d(cout)
+    conv_g = torch.nn.Conv2d(cin, cout, 3, padding=1, bias=False).cuda().to(dtype)
+    bn_g = torch.nn.BatchNorm2d(cout).cuda()  # BN affine/stats stay fp32
+    conv_g.load_state_dict({k: v.to(dtype) for k, v in conv.state_dict().items()})
+    bn_g.load_state_dict(bn.state_dict())
+    bn.train(training), bn_g.train(training)
+
+    x = torch.randn(n, cin, hw, hw)
+    xg = x.cuda().to(dtype).contiguous(memory_format=CL).requires_grad_(True)
+    xr = x.clone().requires_grad_(True)
+
+    y = ops.conv_bn_act(xg, conv_g, bn_g, act=True, training=training)
+    yr = F.leaky_relu(bn(conv(xr)), 0.01)
+    _assert_close(y, yr, tol, tol, "fwd")
+
+    dy = torch.randn_like(yr)
+    yr.backward(dy)
+    y.backward(dy.cuda().to(dtype))
+    _assert_close(xg.grad, xr.grad, tol * 4, tol * 4, "dx")
+    _assert_close(conv_g.weight.grad, conv.weight.grad, tol * 4,
+                  tol * 4 * conv.weight.grad.abs().max().item() + tol, "dw")
+    _assert_close(bn_g.weight.grad, bn.weight.grad, tol * 4, tol * 4, "dgamma")
+    _assert_close(bn_g.bias.grad, bn.bias.grad, tol * 4, tol * 4, "dbeta")
+    if training:
+        _assert_close(bn_g.running_mean, bn.running_mean, tol, tol, "running_mean")
+        _assert_close(bn_g.running_var, bn.running_var, tol, tol, "running_var")
+
+
+def test_conv_bn_add_act_residual():
+    torch.manual_seed(1)
+    conv = torch.nn.Conv2d(8, 8, 1, bias=False)
+    bn = torch.nn.BatchNorm2d(8)
+    conv_g = torch.nn.Conv2d(8, 8, 1, bias=False).cuda()
+    bn_g = torch.nn.BatchNorm2d(8).cuda()
+    conv_g.load_state_dict(conv.state_dict())
+    bn_g.load_state_dict(bn.state_dict())
+    x = torch.randn(2, 8, 8, 8)
+    r = torch.randn(2, 8, 8, 8)
+    xg = x.cuda().contiguous(memory_format=CL).requires_grad_(True)
+    rg = r.cuda().contiguous(memory_format=CL).requires_grad_(True)
+    xr = x.clone().requires_grad_(True)
+    rr = r.clone().requires_grad_(True)
+    y = ops.conv_bn_add_act(xg, conv_g, bn_g, rg, act=True, training=True)
+    yr = F.leaky_relu(bn(conv(xr)) + rr, 0.01)
+    _assert_close(y, yr, 1e-4, 1e-4, "fwd")
+    dy = torch.randn_like(yr)
+    yr.backward(dy)
+    y.backward(dy.cuda())
+    _assert_close(rg.grad, rr.grad, 1e-4, 1e-4, "dres")
+    _assert_close(xg.grad, xr.grad, 1e-3, 1e-4, "dx")
+
+
+def test_mfma_conv_shapes_vs_miopen():
+    """MFMA implicit-GEMM conv vs the library conv across the model's shapes."""
+    from improved_body_parts_amd.ops import conv_kernels
+    ext = _backend.hip_extension()
+    if not hasattr(ext, "conv_mfma_fwd"):
+        pytest.skip("MFMA conv not built yet")
+    cases = [
+        (2, 64, 64, 64, 1, 1, 1),     # 1x1
+        (2, 64, 128, 32, 3, 1, 1),    # 3x3
+        (1, 128, 128, 128, 3, 1, 3),  # dilated 3
+        (1, 128, 128, 64, 3, 1, 5),   # dilated 5
+        (2, 3, 64, 64, 7, 2, 1),      # stem 7x7 s2
+        (2, 384, 384, 8, 3, 1, 1),    # small spatial, wide channels
+        (2, 256, 50, 32, 1, 1, 1),    # head 1x1 to 50ch
+    ]
+    for n, cin, cout, hw, k, s, d in cases:
+        torch.manual_seed(0)
+        x = torch.randn(n, cin, hw, hw, device="cuda").bfloat16() \
+            .contiguous(memory_format=CL)
+        w = (torch.randn(cout, cin, k, k, device="cuda") * 0.05).bfloat16()
+        pad = (k - 1) // 2 * d
+        y = conv_kernels.conv_fwd(x, w, (s, s), (pad, pad), (d, d))
+        assert y is not None, f"shape not covered: {(n, cin, cout, hw, k, s, d)}"
+        ref = F.conv2d(x.float(), w.float(), None, s, pad, d)
+        _assert_close(y, ref, 5e-2, 5e-2 * ref.abs().max().item(),
+                      f"conv {(n, cin, cout, hw, k, s, d)}")
+
+
+# ---------------------------------------------------------------------------
+# spatial ops
+# ---------------------------------------------------------------------------
+def test_maxpool2x2():
+    x = torch.randn(2, 32, 16, 16).cuda().contiguous(memory_format=CL)
+    xg = x.requires_grad_(True)
+    y = ops.maxpool2x2(xg)
+    ref_in = x.detach().clone().cpu().requires_grad_(True)
+    yr = F.max_pool2d(ref_in, 2, 2)
+    _assert_close(y, yr, 1e-6, 1e-6)
+    dy = torch.randn_like(yr)
+    yr.backward(dy)
+    y.backward(dy.cuda())
+    _assert_close(xg.grad, ref_in.grad, 1e-6, 1e-6, "maxpool dx")
+
+
+def test_upsample2x():
+    x = torch.randn(2, 16, 8, 8).cuda().contiguous(memory_format=CL)
+    xg = x.requires_grad_(True)
+    y = ops.upsample2x_nearest(xg)
+    ref_in = x.detach().clone().cpu().requires_grad_(True)
+    yr = F.interpolate(ref_in, scale_factor=2, mode="nearest")
+    _assert_close(y, yr, 1e-6, 1e-6)
+    dy = torch.randn_like(yr)
+    yr.backward(dy)
+    y.backward(dy.cuda())
+    _assert_close(xg.grad, ref_in.grad, 1e-6, 1e-6, "upsample dx")
+
+
+def test_se_layer():
+    torch.manual_seed(3)
+    fc1 = torch.nn.Linear(32, 2)
+    fc2 = torch.nn.Linear(2, 32)
+    fc1g = torch.nn.Linear(32, 2).cuda()
+    fc2g = torch.nn.Linear(2, 32).cuda()
+    fc1g.load_state_dict(fc1.state_dict())
+    fc2g.load_state_dict(fc2.state_dict())
+    x = torch.randn(2, 32, 8, 8)
+    xg = x.cuda().contiguous(memory_format=CL).requires_grad_(True)
+    xr = x.clone().requires_grad_(True)
+    y = ops.se_layer(xg, fc1g, fc2g)
+    n, c = xr.shape[:2]
+    pr = xr.mean(dim=(2, 3))
+    sr = torch.sigmoid(fc2(F.leaky_relu(fc1(pr), 0.01)))
+    yr = xr * sr.view(n, c, 1, 1)
+    _assert_close(y, yr, 1e-4, 1e-5, "se fwd")
+    dy = torch.randn_like(yr)
+    yr.backward(dy)
+    y.backward(dy.cuda())
+    _assert_close(xg.grad, xr.grad, 1e-4, 1e-5, "se dx")
+    _assert_close(fc1g.weight.grad, fc1.weight.grad, 1e-3, 1e-5, "se dfc1")
+
+
+# ---------------------------------------------------------------------------
+# focal loss
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("gamma", [1, 2])
+def test_focal_l2_gpu_vs_cpu(gamma):
+    torch.manual_seed(0)
+    pred = torch.rand(2, 2, 50, 16, 16)
+    gt = torch.rand(2, 50, 16, 16) * (torch.rand(2, 50, 16, 16) > 0.6)
+    mask = (torch.rand(2, 1, 16, 16) > 0.2).float()
+    kw = dict(heat_start=30, bkg_start=48, gamma=gamma,
+              multi_task_weight=0.1, keypoint_task_weight=3.0,
+              nstack_weight=(1, 2))
+    pg = pred.cuda().requires_grad_(True)
+    pr = pred.clone().requires_grad_(True)
+    lg = ops.focal_l2_loss(pg, gt.cuda(), mask.cuda(), **kw)
+    lr = ops.focal_l2_loss(pr, gt, mask, **kw)
+    _assert_close(lg, lr, 1e-4, 1e-3 * float(lr), "loss")
+    lg.backward()
+    lr.backward()
+    _assert_close(pg.grad, pr.grad, 1e-4, 1e-5, "dpred")
+
+
+def test_focal_l2_bf16():
+    torch.manual_seed(1)
+    pred = torch.rand(1, 2, 50, 32, 32)
+    gt = (torch.rand(2, 50, 32, 32) * (torch.rand(2, 50, 32, 32) > 0.6))
+    mask = torch.ones(2, 1, 32, 32)
+    kw = dict(heat_start=30, bkg_start=48, gamma=1, nstack_weight=(1,))
+    pg = pred.cuda().bfloat16().requires_grad_(True)
+    lr = ops.focal_l2_loss(pred.clone().requires_grad_(True), gt, mask, **kw)
+    lg = ops.focal_l2_loss(pg, gt.cuda().bfloat16(), mask.cuda().bfloat16(), **kw)
+    assert abs(float(lg) - float(lr)) / float(lr) < 0.05
+
+
+# ---------------------------------------------------------------------------
+# fused SGD
+# ---------------------------------------------------------------------------
+def test_fused_sgd_bf16_master_weights():
+    from improved_body_parts_amd.engine import FusedSGD
+    torch.manual_seed(0)
+    ref = torch.nn.Linear(64, 64)
+    dev = torch.nn.Linear(64, 64).cuda().bfloat16()
+    dev.load_state_dict({k: v.bfloat16() for k, v in ref.state_dict().items()})
+    opt_ref = torch.optim.SGD(ref.parameters(), lr=0.05, momentum=0.9,
+                              weight_decay=0.01)
+    opt_dev = FusedSGD(dev.parameters(), lr=0.05, momentum=0.9, weight_decay=0.01)
+    for _ in range(10):
+        g = torch.randn(64, 64) * 0.1
+        gb = torch.randn(64) * 0.1
+        opt_ref.zero_grad(), opt_dev.zero_grad()
+        ref.weight.grad = g.clone()
+        ref.bias.grad = gb.clone()
+        dev.weight.grad = g.cuda().bfloat16()
+        dev.bias.grad = gb.cuda().bfloat16()
+        opt_ref.step(), opt_dev.step()
+    # master weights track fp32 SGD closely even after 10 bf16 steps
+    master = opt_dev.state[dev.weight]["master"]
+    _assert_close(master, ref.weight.detach(), 2e-2, 2e-2, "sgd master")
+
+
+# ---------------------------------------------------------------------------
+# post-process
+# ---------------------------------------------------------------------------
+def test_heatmap_nms_matches_util():
+    from improved_body_parts_amd.utils import keypoint_heatmap_nms
+    torch.manual_seed(0)
+    heat = torch.rand(1, 18, 64, 64)
+    got = ops.heatmap_nms(heat.cuda(), 0.1)
+    ref = keypoint_heatmap_nms(heat, 3, 0.1)
+    _assert_close(got, ref, 1e-6, 1e-6)
+
+
+def test_find_peaks_device():
+    from improved_body_parts_amd.ops.postproc import find_peaks_device
+    heat = torch.zeros(3, 64, 64)
+    heat[0, 10, 20] = 0.9
+    heat[2, 40, 30] = 0.8
+    # slight spread so centroid refinement has signal
+    heat[0, 10, 21] = 0.5
+    peaks = find_peaks_device(heat.cuda(), threshold=0.3, radius=2)
+    assert peaks.shape[0] == 2
+    rows = {int(r[0]): r for r in peaks}
+    assert 0 in rows and 2 in rows
+    assert abs(float(rows[0][1]) - 20) < 1.0 and abs(float(rows[0][2]) - 10) < 0.5
+    assert abs(float(rows[2][1]) - 30) < 0.5
+
+
+# ---------------------------------------------------------------------------
+# end-to-end: tiny Network on GPU vs CPU oracle
+# ---------------------------------------------------------------------------
+def test_network_gpu_matches_cpu_fp32():
+    from improved_body_parts_amd.config import CanonicalConfig, TrainingOpt
+    from improved_body_parts_amd.models import Network
+    torch.manual_seed(0)
+    cfg = CanonicalConfig(128, 128, 4)
+    opt = TrainingOpt(nstack=2, hourglass_inp_dim=64, increase=32, batch_size=1,
+                      nstack_weight=[1, 1])
+    net = Network(opt, cfg, bn=True, dist=True)
+    net_g = Network(opt, cfg, bn=True, dist=True).cuda()
+    net_g.load_state_dict(net.state_dict())
+    img = torch.rand(1, 128, 128, 3)
+    mm = torch.ones(1, 1, 32, 32)
+    hm = torch.rand(1, 50, 32, 32)
+    net.train(), net_g.train()
+    loss = net((img, mm, hm))
+    loss_g = net_g((img.cuda(), mm.cuda(), hm.cuda()))
+    assert abs(float(loss_g) - float(loss)) / float(loss) < 1e-3
+    loss.backward()
+    loss_g.backward()
+    g_cpu = net.posenet.pre.conv1.weight.grad
+    g_gpu = net_g.posenet.pre.conv1.weight.grad
+    _assert_close(g_gpu, g_cpu, 1e-3, 1e-3 * g_cpu.abs().max().item() + 1e-6,
+                  "e2e conv1 grad")
+
+
+def test_network_bf16_trains():
+    from improved_body_parts_amd.config import CanonicalConfig, TrainingOpt
+    from improved_body_parts_amd.data import SyntheticPoseDataset
+    from improved_body_parts_amd.engine import FusedSGD
+    from improved_body_parts_amd.models import Network
+    torch.manual_seed(0)
+    cfg = CanonicalConfig(128, 128, 4)
+    opt = TrainingOpt(nstack=2, hourglass_inp_dim=64, increase=32, batch_size=2,
+                      nstack_weight=[1, 1])
+    net = Network(opt, cfg, bn=True, dist=True).cuda().bfloat16()
+    for m in net.modules():
+        if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+            m.float()
+    sgd = FusedSGD(net.parameters(), lr=1e-4, momentum=0.9)
+    ds = SyntheticPoseDataset(cfg, length=2)
+    img, mm, hm = ds[0]
+    batch = tuple(t[None].cuda().bfloat16() for t in (img, mm, hm))
+    losses = []
+    for _ in range(5):
+        sgd.zero_grad()
+        loss = net(batch)
+        loss.backward()
+        sgd.step()
+        losses.append(float(loss))
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0]
