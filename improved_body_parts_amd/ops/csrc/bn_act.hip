@@ -150,12 +150,16 @@ __global__ void bn_act_bwd_apply_kernel(
 // ===========================================================================
 using torch::Tensor;
 
+static inline bool dense_ok(const Tensor& t) {
+  return t.is_contiguous() || t.is_contiguous(at::MemoryFormat::ChannelsLast);
+}
+
 static inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
 std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
-  TORCH_CHECK(x_mc.is_cuda() && x_mc.is_contiguous());
+  TORCH_CHECK(x_mc.is_cuda() && dense_ok(x_mc));
   long long M = x_mc.numel() / C;
   auto opts = x_mc.options().dtype(torch::kFloat32);
   Tensor sums = torch::zeros({C}, opts);
@@ -175,13 +179,13 @@ std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
 
 Tensor bn_act_fwd(const Tensor& x, const Tensor& scale, const Tensor& shift,
                   const c10::optional<Tensor>& residual, double slope, bool act) {
-  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.is_cuda() && dense_ok(x));
   int C = (int)scale.numel();
   long long total = x.numel();
   Tensor y = torch::empty_like(x);
   const void* resp = nullptr;
   if (residual.has_value()) {
-    TORCH_CHECK(residual->is_contiguous() && residual->numel() == total);
+    TORCH_CHECK(dense_ok(*residual) && residual->numel() == total);
     resp = residual->data_ptr();
   }
   dim3 block(256);
@@ -215,7 +219,7 @@ std::vector<Tensor> bn_act_bwd(const Tensor& dy, const Tensor& y, const Tensor& 
                                const c10::optional<Tensor>& mean,
                                const c10::optional<Tensor>& invstd,
                                double slope, bool act, bool need_xhat, int64_t C) {
-  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(dy.is_cuda() && dense_ok(dy) && dense_ok(y));
   long long M = dy.numel() / C;
   auto fopts = dy.options().dtype(torch::kFloat32);
   Tensor dpre = torch::empty_like(dy);

@@ -91,11 +91,12 @@ class _SeScaleFn(torch.autograd.Function):
         x = x.contiguous(memory_format=_CL)
         n, c, h, w = x.shape
         s32 = s.float().contiguous()
+        # se_scale allocates like x (NCHW shape, channels_last strides)
         y = ext.se_scale(x, s32, None, n, h * w, c)
         ctx.save_for_backward(x, s32)
         ctx.dims = (n, h, w, c)
         ctx.sdtype = s.dtype
-        return y.permute(0, 3, 1, 2)
+        return y
 
     @staticmethod
     def backward(ctx, dy):
@@ -105,7 +106,7 @@ class _SeScaleFn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=_CL)
         dx = ext.se_scale(dy, s32, None, n, h * w, c)
         ds = ext.se_reduce(dy, x, n, h * w, c)
-        return dx.permute(0, 3, 1, 2), ds.to(ctx.sdtype)
+        return dx, ds.to(ctx.sdtype)
 
 
 def se_layer_hip(x, fc1, fc2):
