@@ -30,6 +30,9 @@ def _supported(x, weight, stride, padding, dilation, for_grad=False):
     if x.dtype != torch.bfloat16:
         return False
     kh, kw = weight.shape[2], weight.shape[3]
+    cin = x.shape[1]
+    if (kh, kw) != (1, 1) and cin % 64 != 0:
+        return False  # KxK path needs BK | Cin (the 7x7 Cin=3 stem falls back)
     if padding[0] != (kh - 1) // 2 * dilation[0] or \
        padding[1] != (kw - 1) // 2 * dilation[1]:
         return False
@@ -44,21 +47,22 @@ _pack_cache = {}
 
 
 def packed_weight(weight: torch.Tensor) -> torch.Tensor:
-    """[Cout, Cin, kh, kw] -> bf16 [kh*kw*Cin, Cout] contiguous, cached."""
+    """[Cout, Cin, kh, kw] -> bf16 [Cout][kh*kw*Cin] contiguous (N-major for
+    the kernel's B-tile row loads), cached against the version counter."""
     key = id(weight)
     entry = _pack_cache.get(key)
     ver = weight._version
     if entry is not None and entry[0] == ver:
         return entry[1]
     w = weight.detach().to(torch.bfloat16)
-    packed = w.permute(2, 3, 1, 0).reshape(-1, weight.shape[0]).contiguous()
+    packed = w.permute(0, 2, 3, 1).reshape(weight.shape[0], -1).contiguous()
     _pack_cache[key] = (ver, packed)
     return packed
 
 
 def packed_weight_dgrad(weight: torch.Tensor) -> torch.Tensor:
-    """Weights for dgrad-as-conv: rotate 180° spatially and swap Cin/Cout ->
-    [kh*kw*Cout, Cin]."""
+    """Weights for dgrad-as-conv: rotate 180° spatially, swap Cin/Cout ->
+    [Cin][kh*kw*Cout]."""
     key = (id(weight), "dgrad")
     entry = _pack_cache.get(key)
     ver = weight._version
@@ -66,7 +70,7 @@ def packed_weight_dgrad(weight: torch.Tensor) -> torch.Tensor:
         return entry[1]
     w = weight.detach().to(torch.bfloat16)
     w = torch.flip(w, dims=(2, 3))          # rotate the kernel
-    packed = w.permute(2, 3, 0, 1).reshape(-1, weight.shape[1]).contiguous()
+    packed = w.permute(1, 2, 3, 0).reshape(weight.shape[1], -1).contiguous()
     _pack_cache[key] = (ver, packed)
     return packed
 

@@ -37,6 +37,15 @@ Tensor focal_l2_bwd(const Tensor& pred, const Tensor& gt, const Tensor& mask,
                     int64_t bkg_start, int64_t gamma, double mtw, double ktw,
                     double alpha, double beta);
 
+// conv_mfma.hip
+Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
+                     int64_t H, int64_t W, int64_t Cin, int64_t Cout,
+                     int64_t KH, int64_t KW, int64_t stride, int64_t pad_h,
+                     int64_t pad_w, int64_t dil_h, int64_t dil_w, int64_t Ho,
+                     int64_t Wo, const c10::optional<Tensor>& scale,
+                     const c10::optional<Tensor>& shift,
+                     const c10::optional<Tensor>& residual, bool act);
+
 // sgd.hip
 void fused_sgd(const Tensor& chunk_table, double lr, double momentum,
                double weight_decay, int64_t dtype_tag);
@@ -61,6 +70,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("se_scale", &se_scale);
   m.def("focal_l2_fwd", &focal_l2_fwd);
   m.def("focal_l2_bwd", &focal_l2_bwd);
+  m.def("conv_mfma_fwd", &conv_mfma_fwd, "MFMA implicit-GEMM conv forward",
+        py::arg("x"), py::arg("w_packed"), py::arg("N"), py::arg("H"),
+        py::arg("W"), py::arg("Cin"), py::arg("Cout"), py::arg("KH"),
+        py::arg("KW"), py::arg("stride"), py::arg("pad_h"), py::arg("pad_w"),
+        py::arg("dil_h"), py::arg("dil_w"), py::arg("Ho"), py::arg("Wo"),
+        py::arg("scale") = c10::nullopt, py::arg("shift") = c10::nullopt,
+        py::arg("residual") = c10::nullopt, py::arg("act") = false);
   m.def("fused_sgd", &fused_sgd);
   m.def("heatmap_nms", &heatmap_nms);
   m.def("collect_peaks", &collect_peaks);

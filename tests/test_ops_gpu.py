@@ -111,9 +111,12 @@ def test_mfma_conv_shapes_vs_miopen():
         (2, 64, 128, 32, 3, 1, 1),    # 3x3
         (1, 128, 128, 128, 3, 1, 3),  # dilated 3
         (1, 128, 128, 64, 3, 1, 5),   # dilated 5
-        (2, 3, 64, 64, 7, 2, 1),      # stem 7x7 s2
         (2, 384, 384, 8, 3, 1, 1),    # small spatial, wide channels
         (2, 256, 50, 32, 1, 1, 1),    # head 1x1 to 50ch
+        (1, 50, 256, 32, 1, 1, 1),    # merge 1x1 from 50ch (K-tail)
+        (2, 64, 64, 64, 1, 2, 1),     # 1x1 stride 2
+        (1, 256, 77, 16, 1, 1, 1),    # odd Cout tail
+        (3, 192, 320, 20, 3, 1, 1),   # odd M tail (3*20*20=1200 pixels)
     ]
     for n, cin, cout, hw, k, s, d in cases:
         torch.manual_seed(0)
@@ -124,8 +127,30 @@ def test_mfma_conv_shapes_vs_miopen():
         y = conv_kernels.conv_fwd(x, w, (s, s), (pad, pad), (d, d))
         assert y is not None, f"shape not covered: {(n, cin, cout, hw, k, s, d)}"
         ref = F.conv2d(x.float(), w.float(), None, s, pad, d)
-        _assert_close(y, ref, 5e-2, 5e-2 * ref.abs().max().item(),
-                      f"conv {(n, cin, cout, hw, k, s, d)}")
+        _assert_rel(y, ref, 1e-2, f"conv {(n, cin, cout, hw, k, s, d)}")
+
+
+def test_mfma_conv_dgrad_vs_reference():
+    from improved_body_parts_amd.ops import conv_kernels
+    ext = _backend.hip_extension()
+    if not hasattr(ext, "conv_mfma_fwd"):
+        pytest.skip("MFMA conv not built yet")
+    for (cin, cout, hw, k, d) in [(64, 128, 32, 3, 1), (128, 128, 16, 3, 3),
+                                  (64, 50, 16, 1, 1), (50, 64, 16, 1, 1)]:
+        torch.manual_seed(1)
+        dy = torch.randn(2, cout, hw, hw, device="cuda").bfloat16() \
+            .contiguous(memory_format=CL)
+        w = (torch.randn(cout, cin, k, k, device="cuda") * 0.05).bfloat16()
+        pad = (k - 1) // 2 * d
+        dx = conv_kernels.conv_dgrad(dy, w, (2, cin, hw, hw), (1, 1),
+                                     (pad, pad), (d, d))
+        if (k, k) != (1, 1) and cout % 64 != 0:
+            assert dx is None
+            continue
+        assert dx is not None
+        ref = torch.nn.grad.conv2d_input((2, cin, hw, hw), w.float(),
+                                         dy.float(), 1, pad, d)
+        _assert_rel(dx, ref, 1e-2, f"dgrad {(cin, cout, hw, k, d)}")
 
 
 # ---------------------------------------------------------------------------
