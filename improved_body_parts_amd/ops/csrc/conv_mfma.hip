@@ -129,7 +129,10 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
       bool inside = a_valid_row && hi >= 0 && hi < p.H && wi >= 0 && wi < p.W;
       const unsigned short* src =
           p.x + (((long long)a_n * p.H + hi) * p.W + wi) * p.Cin + ci;
-      if (inside && ci + 32 <= p.Cin && kk + 32 <= p.K) {
+      // 16-B vector loads need a 16-B-aligned source (odd Cin, e.g. the
+      // 50-channel merge input, makes pixel rows only 2-B aligned)
+      bool a_aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
+      if (inside && a_aligned && ci + 32 <= p.Cin && kk + 32 <= p.K) {
         #pragma unroll
         for (int v = 0; v < 4; ++v)
           *reinterpret_cast<ushortv8*>(&a_reg[v * 8]) =
@@ -153,7 +156,8 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
       int kk = k0 + b_off;
       bool ok = col < p.Cout;
       const unsigned short* src = p.w + (long long)col * p.K + kk;
-      if (ok && kk + b_elems <= p.K) {
+      bool b_aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
+      if (ok && b_aligned && kk + b_elems <= p.K) {
         #pragma unroll
         for (int v = 0; v < 4; ++v)
           if (v * 8 < b_elems)
