@@ -63,6 +63,37 @@ class ConvBnActFn(torch.autograd.Function):
                 stride, padding, dilation, act, training, bn_mod):
         ext = hip_extension()
         x = _to_cl(x)
+
+        # inference fast path: fold BN into the conv epilogue -> ONE kernel
+        if (not training and not torch.is_grad_enabled()
+                and x.dtype == torch.bfloat16):
+            from . import conv_kernels
+            if gamma is not None:
+                # cache the folded scale/shift on the BN module
+                ver = (gamma._version, bn_mod.running_mean._version,
+                       bn_mod.running_var._version)
+                cached = getattr(bn_mod, "_ibp_folded", None)
+                if cached is not None and cached[0] == ver:
+                    scale, shift = cached[1], cached[2]
+                else:
+                    invstd = torch.rsqrt(bn_mod.running_var.float() + bn_mod.eps)
+                    scale = gamma.float() * invstd
+                    shift = beta.float() - bn_mod.running_mean.float() * scale
+                    bn_mod._ibp_folded = (ver, scale, shift)
+            elif bias is not None:
+                scale = torch.ones(weight.shape[0], device=x.device,
+                                   dtype=torch.float32)
+                shift = bias.float()
+            else:
+                scale = shift = None
+            if scale is not None or act or residual is not None:
+                y = conv_kernels.conv_fwd(x, weight, stride, padding, dilation,
+                                          scale=scale, shift=shift,
+                                          residual=residual, act=act)
+                if y is not None:
+                    ctx.conf = None
+                    return y
+
         y_conv = _conv_forward(x, weight, bias if gamma is None else None,
                                stride, padding, dilation)
         C = y_conv.shape[1]

@@ -43,39 +43,46 @@ def _supported(x, weight, stride, padding, dilation, for_grad=False):
     return True
 
 
+import weakref
+
 _pack_cache = {}
 
 
-def packed_weight(weight: torch.Tensor) -> torch.Tensor:
-    """[Cout, Cin, kh, kw] -> bf16 [Cout][kh*kw*Cin] contiguous (N-major for
-    the kernel's B-tile row loads), cached against the version counter."""
-    key = id(weight)
+def _cached_pack(weight: torch.Tensor, tag: str, pack_fn) -> torch.Tensor:
+    """Cache packs per weight tensor, invalidated by the autograd version
+    counter; a weakref finalizer drops the entry when the weight dies (a bare
+    id() key would alias recycled ids)."""
+    key = (id(weight), tag)
     entry = _pack_cache.get(key)
     ver = weight._version
     if entry is not None and entry[0] == ver:
         return entry[1]
-    w = weight.detach().to(torch.bfloat16)
-    packed = w.permute(0, 2, 3, 1).reshape(weight.shape[0], -1).contiguous()
+    packed = pack_fn(weight.detach().to(torch.bfloat16))
+    if entry is None:
+        weakref.finalize(weight, _pack_cache.pop, key, None)
     _pack_cache[key] = (ver, packed)
     return packed
+
+
+def packed_weight(weight: torch.Tensor) -> torch.Tensor:
+    """[Cout, Cin, kh, kw] -> bf16 [Cout][kh*kw*Cin] contiguous (N-major for
+    the kernel's B-tile row loads)."""
+    return _cached_pack(weight, "fwd", lambda w: w.permute(0, 2, 3, 1)
+                        .reshape(w.shape[0], -1).contiguous())
 
 
 def packed_weight_dgrad(weight: torch.Tensor) -> torch.Tensor:
     """Weights for dgrad-as-conv: rotate 180° spatially, swap Cin/Cout ->
     [Cin][kh*kw*Cout]."""
-    key = (id(weight), "dgrad")
-    entry = _pack_cache.get(key)
-    ver = weight._version
-    if entry is not None and entry[0] == ver:
-        return entry[1]
-    w = weight.detach().to(torch.bfloat16)
-    w = torch.flip(w, dims=(2, 3))          # rotate the kernel
-    packed = w.permute(1, 2, 3, 0).reshape(weight.shape[1], -1).contiguous()
-    _pack_cache[key] = (ver, packed)
-    return packed
+    return _cached_pack(weight, "dgrad", lambda w: torch.flip(w, dims=(2, 3))
+                        .permute(1, 2, 3, 0).reshape(w.shape[1], -1).contiguous())
 
 
-def conv_fwd(x, weight, stride, padding, dilation):
+def conv_fwd(x, weight, stride, padding, dilation,
+             scale=None, shift=None, residual=None, act=False):
+    """MFMA conv forward; with scale/shift/residual/act set, the folded-BN
+    (+residual +leaky) epilogue runs inside the conv kernel — the whole
+    Conv+BN+LeakyReLU module is ONE kernel on the inference path."""
     if not _supported(x, weight, stride, padding, dilation):
         return None
     ext = hip_extension()
@@ -86,9 +93,12 @@ def conv_fwd(x, weight, stride, padding, dilation):
     cout, _, kh, kw = weight.shape
     ho = (h + 2 * padding[0] - dilation[0] * (kh - 1) - 1) // stride[0] + 1
     wo = (w_ + 2 * padding[1] - dilation[1] * (kw - 1) - 1) // stride[1] + 1
+    if residual is not None:
+        residual = residual.contiguous(memory_format=_CL)
     y = ext.conv_mfma_fwd(x, packed_weight(weight), n, h, w_, cin, cout,
                           kh, kw, stride[0], padding[0], padding[1],
-                          dilation[0], dilation[1], ho, wo)
+                          dilation[0], dilation[1], ho, wo,
+                          scale, shift, residual, act)
     return y.permute(0, 3, 1, 2)  # NHWC buffer -> NCHW view (channels_last)
 
 
