@@ -47,6 +47,8 @@ struct ConvParams {
   int K;                     // KH*KW*Cin
   int n_mtiles;              // ceil(M/BM)
   int act;                   // leaky-relu on epilogue
+  int splitk;                // K-dimension split factor (small-grid layers)
+  float* ws;                 // fp32 workspace for split-K partial accumulation
 };
 
 // LDS tile addressing: unpadded 128-B rows with a T2 XOR swizzle — the 16-lane
@@ -66,11 +68,13 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  // tile coordinates (column-major over mtiles for some L2 friendliness)
+  // tile coordinates; with split-K, a tile's K-slices are adjacent block ids
   const int ntiles_n = (p.Cout + BN - 1) / BN;
   int bid = blockIdx.x;
-  const int mt = bid / ntiles_n;
-  const int nt = bid % ntiles_n;
+  const int sk = bid % p.splitk;
+  const int tile = bid / p.splitk;
+  const int mt = tile / ntiles_n;
+  const int nt = tile % ntiles_n;
   const long long m0 = (long long)mt * BM;
   const int n0 = nt * BN;
 
@@ -110,7 +114,11 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   const int hi_base = a_ho * p.stride - p.pad_h;
   const int wi_base = a_wo * p.stride - p.pad_w;
 
-  const int nk = (p.K + BK - 1) / BK;
+  const int nk_total = (p.K + BK - 1) / BK;
+  const int nk_chunk = (nk_total + p.splitk - 1) / p.splitk;
+  const int k_begin = sk * nk_chunk;
+  const int nk = min(nk_chunk, nk_total - k_begin);
+  if (nk <= 0) return;
 
   unsigned short a_reg[32];
   unsigned short b_reg[32];
@@ -213,16 +221,15 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   };
 
   // ---- main loop: register-staged double buffer ----------------------------
-  load_chunk(0);
+  load_chunk(k_begin);
   write_chunk(0);
   __syncthreads();
   for (int t = 0; t < nk; ++t) {
-    if (t + 1 < nk) load_chunk(t + 1);
+    if (t + 1 < nk) load_chunk(k_begin + t + 1);  // loads hide under the MFMAs
     compute(t & 1);
-    if (t + 1 < nk) {
-      __syncthreads();          // everyone done reading buf (t+1)&1 last time
-      write_chunk((t + 1) & 1);
-    }
+    // writing buf[(t+1)&1] is safe without a barrier: its last readers were
+    // separated by the end-of-iteration barrier of step t-1
+    if (t + 1 < nk) write_chunk((t + 1) & 1);
     __syncthreads();
   }
 
@@ -243,12 +250,34 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
         long long m = m0 + row_base + i * 16 + erow4 + r;
         if (m >= p.M) continue;
         long long idx = m * p.Cout + col;
+        if (p.splitk > 1) {
+          atomicAdd(&p.ws[idx], acc[i][j][r]);  // combine kernel finishes up
+          continue;
+        }
         float v = acc[i][j][r] * sc + sh;
         if (p.res) v += us2f(p.res[idx]);
         if (p.act) v = leaky(v, 0.01f);
         p.y[idx] = f2us(v);
       }
     }
+  }
+}
+
+// split-K combine: y = act(scale*ws + shift (+res)) -> bf16
+__global__ void splitk_combine_kernel(const float* __restrict__ ws,
+                                      unsigned short* __restrict__ y,
+                                      const float* __restrict__ scale,
+                                      const float* __restrict__ shift,
+                                      const unsigned short* __restrict__ res,
+                                      long long total, int C, int act) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float v = ws[i];
+    if (scale) v = v * scale[c] + shift[c];
+    if (res) v += us2f(res[i]);
+    if (act) v = leaky(v, 0.01f);
+    y[i] = f2us(v);
   }
 }
 
@@ -289,13 +318,35 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
   p.y = reinterpret_cast<unsigned short*>(y.data_ptr());
   p.n_mtiles = (int)((p.M + ibp::BM - 1) / ibp::BM);
   auto stream = at::hip::getCurrentHIPStream().stream();
-  if (Cout > 64) {
-    int nt = (int)((Cout + 127) / 128);
-    dim3 grid(p.n_mtiles * nt), block(256);
+  const int BN = Cout > 64 ? 128 : 64;
+  const int ntiles = p.n_mtiles * (int)((Cout + BN - 1) / BN);
+  const int nk_total = (p.K + ibp::BK - 1) / ibp::BK;
+  // split K on small grids so the 256-CU chip stays filled (~2 blocks/CU)
+  int splitk = 1;
+  if (ntiles < 384 && nk_total > 1) {
+    splitk = std::min((int)nk_total, (384 + ntiles - 1) / ntiles);
+  }
+  p.splitk = splitk;
+  Tensor ws;
+  if (splitk > 1) {
+    ws = torch::zeros({(long long)p.M * Cout},
+                      x.options().dtype(torch::kFloat32));
+    p.ws = ws.data_ptr<float>();
+  } else {
+    p.ws = nullptr;
+  }
+  dim3 grid(ntiles * splitk), block(256);
+  if (BN == 128) {
     hipLaunchKernelGGL(ibp::conv_mfma_kernel<128>, grid, block, 0, stream, p);
   } else {
-    dim3 grid(p.n_mtiles), block(256);
     hipLaunchKernelGGL(ibp::conv_mfma_kernel<64>, grid, block, 0, stream, p);
+  }
+  if (splitk > 1) {
+    long long total = (long long)p.M * Cout;
+    dim3 cgrid(ibp::grid_1d(total, 256, 4096)), cblock(256);
+    hipLaunchKernelGGL(ibp::splitk_combine_kernel, cgrid, cblock, 0, stream,
+                       p.ws, p.y, p.scale, p.shift, p.res, total, (int)Cout,
+                       p.act);
   }
   return y;
 }
