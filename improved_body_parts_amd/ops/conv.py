@@ -60,13 +60,14 @@ class ConvBnActFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, gamma, beta, residual,
-                stride, padding, dilation, act, training, bn_mod):
+                stride, padding, dilation, act, training, bn_mod, inference):
         ext = hip_extension()
         x = _to_cl(x)
 
-        # inference fast path: fold BN into the conv epilogue -> ONE kernel
-        if (not training and not torch.is_grad_enabled()
-                and x.dtype == torch.bfloat16):
+        # inference fast path: fold BN into the conv epilogue -> ONE kernel.
+        # `inference` is computed OUTSIDE apply(): grad mode is always off
+        # inside Function.forward, so torch.is_grad_enabled() can't be used here.
+        if inference and x.dtype == torch.bfloat16:
             from . import conv_kernels
             if gamma is not None:
                 # cache the folded scale/shift on the BN module
@@ -151,7 +152,7 @@ class ConvBnActFn(torch.autograd.Function):
             dw = _conv_wgrad(x, dy, weight.shape, stride, padding, dilation) \
                 if ctx.needs_input_grad[1] else None
             return (dx, dw, None, None, None, None,
-                    None, None, None, None, None, None)
+                    None, None, None, None, None, None, None)
 
         dpre, sum_dpre, sum_dxhat = ext.bn_act_bwd(
             dy, y, y_conv, mean if has_bn else None, invstd if has_bn else None,
@@ -177,7 +178,7 @@ class ConvBnActFn(torch.autograd.Function):
         dw = _conv_wgrad(x, dconv, weight.shape, stride, padding, dilation) \
             if ctx.needs_input_grad[1] else None
         return (dx, dw, dbias, dgamma, dbeta, dres,
-                None, None, None, None, None, None)
+                None, None, None, None, None, None, None)
 
 
 def conv_bn_act_hip(x, conv, bn, act: bool, residual=None, training: bool = False):
@@ -185,7 +186,9 @@ def conv_bn_act_hip(x, conv, bn, act: bool, residual=None, training: bool = Fals
     nn.Conv2d / nn.BatchNorm2d containers and runs the fused function."""
     gamma = bn.weight if bn is not None else None
     beta = bn.bias if bn is not None else None
+    bn_training = training and (bn is not None and bn.training)
+    inference = not bn_training and not torch.is_grad_enabled()
     return ConvBnActFn.apply(
         x, conv.weight, conv.bias, gamma, beta, residual,
         conv.stride, conv.padding, conv.dilation, act,
-        training and (bn is not None and bn.training), bn)
+        bn_training, bn, inference)

@@ -251,7 +251,9 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
         if (m >= p.M) continue;
         long long idx = m * p.Cout + col;
         if (p.splitk > 1) {
-          atomicAdd(&p.ws[idx], acc[i][j][r]);  // combine kernel finishes up
+          // each K-split owns a workspace slice: plain stores, no zero-init,
+          // no atomics, deterministic sums (combine kernel reduces slices)
+          p.ws[(long long)sk * p.M * p.Cout + idx] = acc[i][j][r];
           continue;
         }
         float v = acc[i][j][r] * sc + sh;
@@ -269,11 +271,13 @@ __global__ void splitk_combine_kernel(const float* __restrict__ ws,
                                       const float* __restrict__ scale,
                                       const float* __restrict__ shift,
                                       const unsigned short* __restrict__ res,
-                                      long long total, int C, int act) {
+                                      long long total, int C, int act,
+                                      int splitk) {
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
        i += (long long)gridDim.x * blockDim.x) {
     int c = (int)(i % C);
-    float v = ws[i];
+    float v = 0.f;
+    for (int s = 0; s < splitk; ++s) v += ws[(long long)s * total + i];
     if (scale) v = v * scale[c] + shift[c];
     if (res) v += us2f(res[i]);
     if (act) v = leaky(v, 0.01f);
@@ -329,7 +333,9 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
   p.splitk = splitk;
   Tensor ws;
   if (splitk > 1) {
-    ws = torch::zeros({(long long)p.M * Cout},
+    // per-split slices written with plain stores -> empty() is safe (every
+    // in-range element is covered by every split's epilogue)
+    ws = torch::empty({(long long)splitk * p.M * Cout},
                       x.options().dtype(torch::kFloat32));
     p.ws = ws.data_ptr<float>();
   } else {
@@ -346,7 +352,7 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
     dim3 cgrid(ibp::grid_1d(total, 256, 4096)), cblock(256);
     hipLaunchKernelGGL(ibp::splitk_combine_kernel, cgrid, cblock, 0, stream,
                        p.ws, p.y, p.scale, p.shift, p.res, total, (int)Cout,
-                       p.act);
+                       p.act, splitk);
   }
   return y;
 }

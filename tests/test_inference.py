@@ -1,0 +1,208 @@
+"""Inference pipeline tests: peaks -> connections -> people -> COCO keypoints.
+
+Strategy: the reference has no tests, so these are built as
+oracle/property tests — synthetic Gaussian heatmaps with KNOWN person layouts
+go through the full assignment pipeline and the assembled people must match
+the layout. Runs on CPU (same code path as GPU minus the HIP kernels, which
+tests/test_ops_gpu.py compares against these ops on device).
+"""
+import numpy as np
+import pytest
+import torch
+
+from improved_body_parts_amd.config import GetConfig, InferenceParams, TrainingOpt
+from improved_body_parts_amd.engine.inference import (
+    find_connections, find_peaks, find_people, format_results, predict, process,
+    subsets_to_keypoints)
+from improved_body_parts_amd.models import NetworkEval
+
+
+@pytest.fixture(scope="module")
+def config():
+    return GetConfig("Canonical")
+
+
+@pytest.fixture(scope="module")
+def params():
+    p, mp = InferenceParams().as_params_dict()
+    return p, mp
+
+
+def _synth_maps(config, people, H=128, W=128, sigma=3.0):
+    """Render heatmap/paf planes for a list of people.
+
+    Each person is a dict part_index -> (x, y). Limb channels get a thick
+    Gaussian ridge along each present limb (enough for the line integral)."""
+    n_heat = config.num_layers - config.paf_layers
+    heat = np.zeros((H, W, n_heat), dtype=np.float32)
+    paf = np.zeros((H, W, config.paf_layers), dtype=np.float32)
+    yy, xx = np.mgrid[0:H, 0:W].astype(np.float32)
+    for person in people:
+        for part, (x, y) in person.items():
+            g = np.exp(-((xx - x) ** 2 + (yy - y) ** 2) / (2 * sigma ** 2))
+            heat[..., part] = np.maximum(heat[..., part], g)
+        for k, (a, b) in enumerate(config.limbs_conn):
+            if a in person and b in person:
+                ax, ay = person[a]
+                bx, by = person[b]
+                # distance from each pixel to the segment
+                vx, vy = bx - ax, by - ay
+                L2 = vx * vx + vy * vy + 1e-9
+                t = ((xx - ax) * vx + (yy - ay) * vy) / L2
+                t = np.clip(t, 0, 1)
+                d2 = (xx - (ax + t * vx)) ** 2 + (yy - (ay + t * vy)) ** 2
+                paf[..., k] = np.maximum(paf[..., k], np.exp(-d2 / (2 * sigma ** 2)))
+    return torch.from_numpy(heat), torch.from_numpy(paf)
+
+
+TWO_PEOPLE = [
+    {0: (30.0, 20.0), 1: (30.0, 35.0), 2: (20.0, 35.0), 3: (18.0, 55.0),
+     5: (40.0, 35.0), 6: (42.0, 55.0)},
+    {0: (90.0, 25.0), 1: (90.0, 40.0), 2: (80.0, 40.0), 3: (78.0, 60.0),
+     5: (100.0, 40.0), 6: (102.0, 60.0)},
+]
+
+
+def test_find_peaks_locates_people(config, params):
+    p, _ = params
+    heat, _ = _synth_maps(config, TWO_PEOPLE)
+    all_peaks = find_peaks(heat, p, config)
+    assert len(all_peaks) == config.heat_layers
+    for part in (0, 1, 2, 3, 5, 6):
+        peaks = all_peaks[part]
+        assert len(peaks) == 2, f"part {part}: {peaks}"
+        found = sorted((pk[0], pk[1]) for pk in peaks)
+        want = sorted(person[part] for person in TWO_PEOPLE)
+        for (fx, fy), (wx, wy) in zip(found, want):
+            assert abs(fx - wx) < 1.0 and abs(fy - wy) < 1.0
+    # peak ids must be globally unique and dense
+    ids = [pk[3] for sub in all_peaks for pk in sub]
+    assert sorted(ids) == list(range(len(ids)))
+
+
+def test_find_connections_matches_within_person(config, params):
+    p, _ = params
+    heat, paf = _synth_maps(config, TWO_PEOPLE)
+    all_peaks = find_peaks(heat, p, config)
+    connection_all, special_k = find_connections(all_peaks, paf, 128, p, config)
+    assert len(connection_all) == config.paf_layers
+    flat = [pk for sub in all_peaks for pk in sub]
+    flat.sort(key=lambda q: q[3])
+    for k, (a, b) in enumerate(config.limbs_conn):
+        if a in TWO_PEOPLE[0] and b in TWO_PEOPLE[0]:
+            conns = connection_all[k]
+            assert len(conns) == 2, f"limb {k}: {conns}"
+            for row in conns:
+                pa, pb = flat[int(row[0])], flat[int(row[1])]
+                # endpoints of one connection belong to the same person
+                da = [abs(pa[0] - pp[a][0]) + abs(pa[1] - pp[a][1])
+                      for pp in TWO_PEOPLE]
+                db = [abs(pb[0] - pp[b][0]) + abs(pb[1] - pp[b][1])
+                      for pp in TWO_PEOPLE]
+                assert np.argmin(da) == np.argmin(db)
+
+
+def test_find_people_assembles_two(config, params):
+    p, _ = params
+    heat, paf = _synth_maps(config, TWO_PEOPLE)
+    all_peaks = find_peaks(heat, p, config)
+    connection_all, special_k = find_connections(all_peaks, paf, 128, p, config)
+    subset, candidate = find_people(connection_all, special_k, all_peaks, p, config)
+    assert len(subset) == 2
+    for s in subset:
+        assert s[-1][0] >= 5  # all six parts of a person wired together (neck
+        # limbs may overlap) -> at least 5 assembled parts each
+    kps = subsets_to_keypoints(subset, candidate, config)
+    assert len(kps) == 2
+    for coco, score in kps:
+        assert len(coco) == 17
+        assert 0 < score <= 1
+
+
+def test_find_people_empty_maps(config, params):
+    p, _ = params
+    heat, paf = _synth_maps(config, [])
+    all_peaks = find_peaks(heat, p, config)
+    connection_all, special_k = find_connections(all_peaks, paf, 128, p, config)
+    subset, candidate = find_people(connection_all, special_k, all_peaks, p, config)
+    assert len(subset) == 0
+    assert subsets_to_keypoints(subset, candidate, config) == []
+
+
+def test_merge_disjoint_subsets(config, params):
+    """A person seen as two disconnected halves (one limb missing from the
+    chain) must be merged into one subset when a later limb joins them."""
+    p, _ = params
+    person = {0: (60.0, 20.0), 1: (60.0, 40.0), 2: (45.0, 40.0),
+              5: (75.0, 40.0), 6: (78.0, 62.0), 7: (80.0, 84.0)}
+    heat, paf = _synth_maps(config, [person])
+    all_peaks = find_peaks(heat, p, config)
+    connection_all, special_k = find_connections(all_peaks, paf, 128, p, config)
+    subset, _ = find_people(connection_all, special_k, all_peaks, p, config)
+    assert len(subset) == 1
+    assert subset[0][-1][0] >= 5
+
+
+def test_format_results(tmp_path, config, params):
+    p, _ = params
+    heat, paf = _synth_maps(config, TWO_PEOPLE)
+    all_peaks = find_peaks(heat, p, config)
+    connection_all, special_k = find_connections(all_peaks, paf, 128, p, config)
+    subset, candidate = find_people(connection_all, special_k, all_peaks, p, config)
+    kps = {42: subsets_to_keypoints(subset, candidate, config)}
+    res = format_results(kps, str(tmp_path / "res.json"))
+    assert len(res) == 2
+    for r in res:
+        assert r["image_id"] == 42 and r["category_id"] == 1
+        assert len(r["keypoints"]) == 51
+
+
+def test_predict_shapes_and_flip_consistency(config):
+    """predict() on a small model returns original-resolution maps; a
+    horizontally symmetric input must give (near) symmetric keypoint maps."""
+    opt = TrainingOpt(nstack=1, batch_size=1)
+    model = NetworkEval(opt, config, bn=True).eval()
+    img = np.random.RandomState(0).rand(96, 80, 3).astype(np.float32)
+    p, mp = InferenceParams().as_params_dict()
+    mp = dict(mp)
+    mp["boxsize"] = 96  # keep the search scale at 1 for speed
+    with torch.no_grad():
+        heat, paf = predict(img, model, config, p, mp)
+    assert heat.shape == (96, 80, config.num_layers - config.paf_layers)
+    assert paf.shape == (96, 80, config.paf_layers)
+    assert torch.isfinite(heat).all() and torch.isfinite(paf).all()
+
+
+def test_process_end_to_end(config):
+    opt = TrainingOpt(nstack=1, batch_size=1)
+    model = NetworkEval(opt, config, bn=True).eval()
+    img = np.random.RandomState(1).rand(64, 64, 3).astype(np.float32)
+    p, mp = InferenceParams().as_params_dict()
+    mp = dict(mp)
+    mp["boxsize"] = 64
+    kps = process(img, model, config, p, mp)
+    assert isinstance(kps, list)  # random weights: usually no people, no crash
+
+
+@pytest.mark.gpu
+def test_pipeline_device_vs_cpu_parity(config, params):
+    """Device kernels (NMS + collect_peaks + limb_scores) must reproduce the
+    CPU pipeline on the same synthetic maps."""
+    p, _ = params
+    heat, paf = _synth_maps(config, TWO_PEOPLE)
+    ap_cpu = find_peaks(heat, p, config)
+    cl_cpu, sk_cpu = find_connections(ap_cpu, paf, 128, p, config)
+    sub_cpu, _ = find_people(cl_cpu, sk_cpu, ap_cpu, p, config)
+
+    heat_d, paf_d = heat.cuda(), paf.cuda()
+    ap_dev = find_peaks(heat_d, p, config)
+    cl_dev, sk_dev = find_connections(ap_dev, paf_d, 128, p, config)
+    sub_dev, _ = find_people(cl_dev, sk_dev, ap_dev, p, config)
+
+    assert sk_cpu == sk_dev
+    assert len(sub_cpu) == len(sub_dev) == 2
+    for sc, sd in zip(ap_cpu, ap_dev):
+        assert len(sc) == len(sd)
+        for pc, pd in zip(sorted(sc), sorted(sd)):
+            assert abs(pc[0] - pd[0]) < 0.05 and abs(pc[1] - pd[1]) < 0.05
+            assert abs(pc[2] - pd[2]) < 0.01
