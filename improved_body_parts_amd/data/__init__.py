@@ -3,13 +3,13 @@
 Mirrors the capabilities of the reference's data/ + py_cocodata_server/ packages
 (SURVEY.md §2 L1/L2), with the heatmapper doubled as a HIP kernel on device.
 """
-from .heatmapper import Heatmapper, limb_gaussian
+from .heatmapper import Heatmapper, create_heatmaps_device, limb_gaussian
 from .transformer import Transformer, AugmentSelection
 from .synthetic import SyntheticPoseDataset, sample_people
 from .coco import MyDataset, RawDataIterator, build_coco_h5
 
 __all__ = [
-    "Heatmapper", "limb_gaussian",
+    "Heatmapper", "create_heatmaps_device", "limb_gaussian",
     "Transformer", "AugmentSelection",
     "SyntheticPoseDataset", "sample_people",
     "MyDataset", "RawDataIterator", "build_coco_h5",

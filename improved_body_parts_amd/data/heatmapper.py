@@ -164,6 +164,43 @@ class Heatmapper:
         return off.transpose((2, 0, 1)), mask.transpose((2, 0, 1))
 
 
+def create_heatmaps_device(joints_batch, mask_all_batch, config, device="cuda"):
+    """On-device batched GT generation (HIP kernel csrc/heatmap_gt.hip).
+
+    joints_batch: (N, P, num_parts, 3) array/tensor, original-resolution
+    coordinates, visibility 2 = absent (pad rows with vis=2).
+    mask_all_batch: (N, h, w) stride-grid masks or None.
+    Returns (N, num_layers, h, w) fp32 CUDA tensor — numerically matching the
+    CPU oracle ``Heatmapper.create_heatmaps`` (tested in tests/test_ops_gpu).
+    """
+    import torch as _torch
+
+    from ..ops._backend import require_hip
+
+    ext = require_hip()
+    assert config.paf_start == 0, "device GT kernel assumes PAF-first layout"
+    tp = config.transform_params
+    h = config.height // config.stride
+    w = config.width // config.stride
+    joints = _torch.as_tensor(np.asarray(joints_batch, dtype=np.float32)
+                              if not isinstance(joints_batch, _torch.Tensor)
+                              else joints_batch, dtype=_torch.float32,
+                              device=device)
+    mask = None
+    if mask_all_batch is not None:
+        mask = _torch.as_tensor(np.asarray(mask_all_batch, dtype=np.float32)
+                                if not isinstance(mask_all_batch, _torch.Tensor)
+                                else mask_all_batch, dtype=_torch.float32,
+                                device=device)
+    limb_pairs = _torch.tensor(np.asarray(config.limbs_conn, dtype=np.int32),
+                               dtype=_torch.int32)
+    return ext.heatmap_gt(joints, mask, limb_pairs, h, w, config.stride,
+                          config.heat_start, config.bkg_start,
+                          config.num_layers, tp.sigma, tp.paf_sigma,
+                          tp.keypoint_gaussian_thre, tp.limb_gaussian_thre,
+                          tp.paf_thre)
+
+
 def limb_gaussian(X, Y, sigma, x1, y1, x2, y2, thresh=0.01):
     """Gaussian of the perpendicular distance from grid points to the segment's
     carrier line (reference distances() :309-340, including its quirk of writing

@@ -50,6 +50,13 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
 void fused_sgd(const Tensor& chunk_table, double lr, double momentum,
                double weight_decay, int64_t dtype_tag);
 
+// heatmap_gt.hip
+Tensor heatmap_gt(const Tensor& joints, const c10::optional<Tensor>& mask_all,
+                  const Tensor& limb_pairs, int64_t h, int64_t w,
+                  int64_t stride, int64_t heat_start, int64_t bkg_start,
+                  int64_t num_layers, double sigma, double paf_sigma,
+                  double keypoint_thre, double limb_thre, double paf_thre);
+
 // postproc.hip
 Tensor heatmap_nms(const Tensor& heat, double thre);
 std::vector<Tensor> collect_peaks(const Tensor& nmsed, const Tensor& smoothed,
@@ -78,6 +85,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("scale") = c10::nullopt, py::arg("shift") = c10::nullopt,
         py::arg("residual") = c10::nullopt, py::arg("act") = false);
   m.def("fused_sgd", &fused_sgd);
+  m.def("heatmap_gt", &heatmap_gt, "on-device GT heatmap/PAF generation",
+        py::arg("joints"), py::arg("mask_all"), py::arg("limb_pairs"),
+        py::arg("h"), py::arg("w"), py::arg("stride"), py::arg("heat_start"),
+        py::arg("bkg_start"), py::arg("num_layers"), py::arg("sigma"),
+        py::arg("paf_sigma"), py::arg("keypoint_thre"), py::arg("limb_thre"),
+        py::arg("paf_thre"));
   m.def("heatmap_nms", &heatmap_nms);
   m.def("collect_peaks", &collect_peaks);
   m.def("limb_scores", &limb_scores);
