@@ -40,6 +40,60 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
 // --------------------------------------------------------------------------
 typedef unsigned short ushort8 __attribute__((ext_vector_type(8)));
 
+// --------------------------------------------------------------------------
+// vectorized bf16 stats: each lane owns 8 adjacent channels (one 16-B load
+// per row), a block covers cpb channel-lanes x rpb row-groups, partial sums
+// cross the row-groups through LDS so only cpb lanes touch the atomics.
+// --------------------------------------------------------------------------
+__global__ void bn_stats_bf16v8(const ushort8* __restrict__ x,
+                                float* __restrict__ sums,
+                                float* __restrict__ sumsq,
+                                long long M, int C8, int cpb, int rpb) {
+  __shared__ float red[256 * 8];
+  const int tid = threadIdx.x;
+  const int cl = tid % cpb;
+  const int rs = tid / cpb;
+  const int ch8 = blockIdx.y * cpb + cl;
+  const bool active = ch8 < C8 && rs < rpb;
+  float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float q[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  if (active) {
+    for (long long m = (long long)blockIdx.x * rpb + rs; m < M;
+         m += (long long)gridDim.x * rpb) {
+      ushort8 xv = x[m * C8 + ch8];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = us2f(xv[j]);
+        s[j] += v;
+        q[j] += v * v;
+      }
+    }
+  }
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? s[j] : 0.f;
+  __syncthreads();
+  if (rs == 0 && ch8 < C8) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float acc = 0.f;
+      for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
+      atomicAdd(&sums[ch8 * 8 + j], acc);
+    }
+  }
+  __syncthreads();
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? q[j] : 0.f;
+  __syncthreads();
+  if (rs == 0 && ch8 < C8) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float acc = 0.f;
+      for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
+      atomicAdd(&sumsq[ch8 * 8 + j], acc);
+    }
+  }
+}
+
 __global__ void bn_act_fwd_bf16v8(const unsigned short* __restrict__ x,
                                   const unsigned short* __restrict__ res,
                                   unsigned short* __restrict__ y,
@@ -69,6 +123,108 @@ __global__ void bn_act_fwd_bf16v8(const unsigned short* __restrict__ x,
       }
     }
     yv[i] = out;
+  }
+}
+
+// vectorized bf16 backward reduce: same block geometry as bn_stats_bf16v8;
+// dpre is written back as one 16-B store per row.
+template <bool NEED_XHAT>
+__global__ void bn_act_bwd_reduce_bf16v8(
+    const ushort8* __restrict__ dy, const ushort8* __restrict__ y,
+    const ushort8* __restrict__ x, ushort8* __restrict__ dpre_out,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ sum_dpre, float* __restrict__ sum_dxhat,
+    long long M, int C8, int cpb, int rpb, float slope, int act) {
+  __shared__ float red[256 * 8];
+  const int tid = threadIdx.x;
+  const int cl = tid % cpb;
+  const int rs = tid / cpb;
+  const int ch8 = blockIdx.y * cpb + cl;
+  const bool active = ch8 < C8 && rs < rpb;
+  float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float sx[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float mu[8], is[8];
+  if (active && NEED_XHAT) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mu[j] = mean[ch8 * 8 + j];
+      is[j] = invstd[ch8 * 8 + j];
+    }
+  }
+  if (active) {
+    for (long long m = (long long)blockIdx.x * rpb + rs; m < M;
+         m += (long long)gridDim.x * rpb) {
+      const long long i = m * C8 + ch8;
+      ushort8 gd = dy[i];
+      ushort8 yy;
+      if (act) yy = y[i];
+      ushort8 xx;
+      if (NEED_XHAT) xx = x[i];
+      ushort8 out;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = us2f(gd[j]);
+        if (act) g = us2f(yy[j]) > 0.f ? g : g * slope;
+        out[j] = f2us(g);
+        s[j] += g;
+        if (NEED_XHAT) sx[j] += g * (us2f(xx[j]) - mu[j]) * is[j];
+      }
+      dpre_out[i] = out;
+    }
+  }
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? s[j] : 0.f;
+  __syncthreads();
+  if (rs == 0 && ch8 < C8) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float acc = 0.f;
+      for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
+      atomicAdd(&sum_dpre[ch8 * 8 + j], acc);
+    }
+  }
+  if (NEED_XHAT) {
+    __syncthreads();
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? sx[j] : 0.f;
+    __syncthreads();
+    if (rs == 0 && ch8 < C8) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float acc = 0.f;
+        for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
+        atomicAdd(&sum_dxhat[ch8 * 8 + j], acc);
+      }
+    }
+  }
+}
+
+// vectorized bf16 BN backward apply (elementwise, 8 channels per lane)
+__global__ void bn_act_bwd_apply_bf16v8(
+    const ushort8* __restrict__ dpre, const ushort8* __restrict__ x,
+    ushort8* __restrict__ dx, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ sum_dpre, const float* __restrict__ sum_dxhat,
+    long long total8, int C8, float invM) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total8;
+       i += (long long)gridDim.x * blockDim.x) {
+    const int c8 = (int)(i % C8) * 8;
+    ushort8 g8 = dpre[i];
+    ushort8 x8;
+    if (sum_dpre) x8 = x[i];
+    ushort8 out;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c8 + j;
+      float g = us2f(g8[j]);
+      const float is = invstd[c];
+      if (sum_dpre) {
+        float xhat = (us2f(x8[j]) - mean[c]) * is;
+        g = g - sum_dpre[c] * invM - xhat * sum_dxhat[c] * invM;
+      }
+      out[j] = f2us(gamma[c] * is * g);
+    }
+    dx[i] = out;
   }
 }
 
@@ -158,13 +314,37 @@ static inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
+// pick the (channel-lanes, row-groups, row-blocks) geometry for the v8 kernels
+static inline void v8_geometry(long long M, int C8, int& cpb, int& rpb,
+                               int& rows) {
+  cpb = std::min(C8, 256);
+  rpb = 256 / cpb;
+  // >=4 rows of work per thread, capped so the atomic traffic stays small
+  rows = (int)std::min<long long>((M + (long long)rpb * 4 - 1) / (rpb * 4), 1024);
+  rows = std::max(rows, 1);
+}
+
 std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
   TORCH_CHECK(x_mc.is_cuda() && dense_ok(x_mc));
   long long M = x_mc.numel() / C;
   auto opts = x_mc.options().dtype(torch::kFloat32);
-  Tensor sums = torch::zeros({C}, opts);
-  Tensor sumsq = torch::zeros({C}, opts);
+  // one allocation + one async memset for both accumulators (a torch::zeros
+  // pair costs two FillFunctor launches per conv layer — visible in rocprof)
+  Tensor both = torch::empty({2, C}, opts);
+  hipMemsetAsync(both.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
+  Tensor sums = both[0];
+  Tensor sumsq = both[1];
   dim3 block(256);
+  if (x_mc.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
+    int C8 = (int)C / 8, cpb, rpb, rows;
+    v8_geometry(M, C8, cpb, rpb, rows);
+    dim3 grid(rows, (C8 + cpb - 1) / cpb);
+    hipLaunchKernelGGL(ibp::bn_stats_bf16v8, grid, block, 0, cur_stream(),
+                       reinterpret_cast<const ibp::ushort8*>(x_mc.data_ptr()),
+                       sums.data_ptr<float>(), sumsq.data_ptr<float>(),
+                       M, C8, cpb, rpb);
+    return {sums, sumsq};
+  }
   int rows = (int)std::min<long long>((M + 63) / 64, 1024);
   dim3 grid(rows, (C + 255) / 256);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
@@ -223,9 +403,31 @@ std::vector<Tensor> bn_act_bwd(const Tensor& dy, const Tensor& y, const Tensor& 
   long long M = dy.numel() / C;
   auto fopts = dy.options().dtype(torch::kFloat32);
   Tensor dpre = torch::empty_like(dy);
-  Tensor sum_dpre = torch::zeros({C}, fopts);
-  Tensor sum_dxhat = torch::zeros({C}, fopts);
+  Tensor both = torch::empty({2, C}, fopts);
+  hipMemsetAsync(both.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
+  Tensor sum_dpre = both[0];
+  Tensor sum_dxhat = both[1];
   dim3 block(256);
+  if (dy.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
+    int C8 = (int)C / 8, cpb, rpb, rows;
+    v8_geometry(M, C8, cpb, rpb, rows);
+    dim3 gridv(rows, (C8 + cpb - 1) / cpb);
+    auto launch = [&](auto need_xhat) {
+      hipLaunchKernelGGL((ibp::bn_act_bwd_reduce_bf16v8<decltype(need_xhat)::value>),
+                         gridv, block, 0, cur_stream(),
+                         reinterpret_cast<const ibp::ushort8*>(dy.data_ptr()),
+                         reinterpret_cast<const ibp::ushort8*>(y.data_ptr()),
+                         reinterpret_cast<const ibp::ushort8*>(x.data_ptr()),
+                         reinterpret_cast<ibp::ushort8*>(dpre.data_ptr()),
+                         need_xhat.value ? mean->data_ptr<float>() : nullptr,
+                         need_xhat.value ? invstd->data_ptr<float>() : nullptr,
+                         sum_dpre.data_ptr<float>(), sum_dxhat.data_ptr<float>(),
+                         M, C8, cpb, rpb, (float)slope, act ? 1 : 0);
+    };
+    if (need_xhat) launch(std::true_type{});
+    else launch(std::false_type{});
+    return {dpre, sum_dpre, sum_dxhat};
+  }
   int rows = (int)std::min<long long>((M + 63) / 64, 1024);
   dim3 grid(rows, (C + 255) / 256);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
@@ -263,6 +465,21 @@ Tensor bn_act_bwd_apply(const Tensor& dpre, const Tensor& x, const Tensor& mean,
   long long M = dpre.numel() / C;
   Tensor dx = torch::empty_like(dpre);
   dim3 block(256);
+  if (dpre.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
+    long long total8 = dpre.numel() / 8;
+    dim3 gridv(ibp::grid_1d(total8, 256, 8192));
+    hipLaunchKernelGGL(ibp::bn_act_bwd_apply_bf16v8, gridv, block, 0,
+                       cur_stream(),
+                       reinterpret_cast<const ibp::ushort8*>(dpre.data_ptr()),
+                       reinterpret_cast<const ibp::ushort8*>(x.data_ptr()),
+                       reinterpret_cast<ibp::ushort8*>(dx.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(),
+                       sum_dpre.has_value() ? sum_dpre->data_ptr<float>() : nullptr,
+                       sum_dxhat.has_value() ? sum_dxhat->data_ptr<float>() : nullptr,
+                       total8, (int)C / 8, 1.0f / (float)M);
+    return dx;
+  }
   dim3 grid(ibp::grid_1d(dpre.numel(), 256, 8192));
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
       dpre.scalar_type(), "bn_act_bwd_apply", [&] {
