@@ -116,3 +116,36 @@ def test_residual_channel_change():
     assert y.shape == (1, 64, 16, 16)
     r2 = Residual(64, 64, bn=True)
     assert not hasattr(r2, "skipConv")
+
+
+@pytest.mark.parametrize("name", ["final", "attention", "light", "independent"])
+def test_variant_forward_backward(name):
+    """Every 5-scale variant runs fwd+bwd on the CPU plumbing config
+    (reference-variant smoke pattern, e.g. posenet_final.py:211-228)."""
+    from improved_body_parts_amd.models import build_posenet
+    torch.manual_seed(0)
+    net = build_posenet(name, nstack=2, inp_dim=64, oup_dim=50, bn=True,
+                        increase=32)
+    out = net(torch.rand(1, 128, 128, 3))
+    assert len(out) == 2 and len(out[0]) == 5
+    for s, t in enumerate(out[0]):
+        assert t.shape == (1, 50, 32 // (2 ** s), 32 // (2 ** s))
+    out[0][0].sum().backward()
+    grads = [p.grad for p in net.parameters() if p.grad is not None]
+    assert len(grads) > 0 and all(torch.isfinite(g).all() for g in grads)
+
+
+def test_ae_variant_forward_backward():
+    from improved_body_parts_amd.models import build_posenet
+    net = build_posenet("ae", nstack=2, inp_dim=64, oup_dim=50, bn=True,
+                        increase=32)
+    out = net(torch.rand(1, 128, 128, 3))
+    assert len(out) == 2
+    assert out[0].shape == (1, 50, 32, 32)
+    out[-1].sum().backward()
+
+
+def test_variant_registry_rejects_unknown():
+    from improved_body_parts_amd.models import build_posenet
+    with pytest.raises(ValueError):
+        build_posenet("nope", 1, 64, 50)
