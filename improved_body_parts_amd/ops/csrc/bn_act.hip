@@ -42,12 +42,15 @@ typedef unsigned short ushort8 __attribute__((ext_vector_type(8)));
 
 // --------------------------------------------------------------------------
 // vectorized bf16 stats: each lane owns 8 adjacent channels (one 16-B load
-// per row), a block covers cpb channel-lanes x rpb row-groups, partial sums
-// cross the row-groups through LDS so only cpb lanes touch the atomics.
+// per row), a block covers cpb channel-lanes x rpb row-groups. Partial sums
+// cross the row-groups through LDS, then each block writes its cpb x 8
+// channel partials to a [2][rows][C] workspace with PLAIN stores; a second
+// tiny kernel reduces the rows. No atomics anywhere: global fp32 atomics
+// serialise per address (~1024 colliding blocks cost a flat ~240 us floor
+// on every call — measured, gpurun_out/bn_v8.txt of 2026-09-13).
 // --------------------------------------------------------------------------
 __global__ void bn_stats_bf16v8(const ushort8* __restrict__ x,
-                                float* __restrict__ sums,
-                                float* __restrict__ sumsq,
+                                float* __restrict__ ws,  // [2][rows][C]
                                 long long M, int C8, int cpb, int rpb) {
   __shared__ float red[256 * 8];
   const int tid = threadIdx.x;
@@ -69,29 +72,37 @@ __global__ void bn_stats_bf16v8(const ushort8* __restrict__ x,
       }
     }
   }
-  #pragma unroll
-  for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? s[j] : 0.f;
-  __syncthreads();
-  if (rs == 0 && ch8 < C8) {
+  const long long C = (long long)C8 * 8;
+  const long long plane = (long long)gridDim.x * C;
+  for (int pass = 0; pass < 2; ++pass) {
+    const float* src = pass == 0 ? s : q;
+    __syncthreads();
     #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float acc = 0.f;
-      for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
-      atomicAdd(&sums[ch8 * 8 + j], acc);
+    for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? src[j] : 0.f;
+    __syncthreads();
+    if (rs == 0 && ch8 < C8) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float acc = 0.f;
+        for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
+        ws[pass * plane + (long long)blockIdx.x * C + ch8 * 8 + j] = acc;
+      }
     }
   }
-  __syncthreads();
-  #pragma unroll
-  for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? q[j] : 0.f;
-  __syncthreads();
-  if (rs == 0 && ch8 < C8) {
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float acc = 0.f;
-      for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
-      atomicAdd(&sumsq[ch8 * 8 + j], acc);
-    }
-  }
+}
+
+// stage 2: out[p][c] = sum_r ws[p][r][c]; one thread per (plane, channel)
+__global__ void colsum_kernel(const float* __restrict__ ws,
+                              float* __restrict__ out, int rows, int C,
+                              int planes) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= C * planes) return;
+  int p = i / C;
+  int c = i - p * C;
+  const float* src = ws + (long long)p * rows * C + c;
+  float acc = 0.f;
+  for (int r = 0; r < rows; ++r) acc += src[(long long)r * C];
+  out[i] = acc;
 }
 
 __global__ void bn_act_fwd_bf16v8(const unsigned short* __restrict__ x,
@@ -133,7 +144,7 @@ __global__ void bn_act_bwd_reduce_bf16v8(
     const ushort8* __restrict__ dy, const ushort8* __restrict__ y,
     const ushort8* __restrict__ x, ushort8* __restrict__ dpre_out,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ sum_dpre, float* __restrict__ sum_dxhat,
+    float* __restrict__ sum_dpre /* ws [2][rows][C] */,
     long long M, int C8, int cpb, int rpb, float slope, int act) {
   __shared__ float red[256 * 8];
   const int tid = threadIdx.x;
@@ -172,57 +183,68 @@ __global__ void bn_act_bwd_reduce_bf16v8(
       dpre_out[i] = out;
     }
   }
-  #pragma unroll
-  for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? s[j] : 0.f;
-  __syncthreads();
-  if (rs == 0 && ch8 < C8) {
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float acc = 0.f;
-      for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
-      atomicAdd(&sum_dpre[ch8 * 8 + j], acc);
-    }
-  }
-  if (NEED_XHAT) {
+  // per-block partials to ws[2][rows][C] (plain stores; colsum finishes)
+  const long long C = (long long)C8 * 8;
+  const long long plane = (long long)gridDim.x * C;
+  for (int pass = 0; pass < 2; ++pass) {
+    const float* src = pass == 0 ? s : sx;
     __syncthreads();
     #pragma unroll
-    for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? sx[j] : 0.f;
+    for (int j = 0; j < 8; ++j) red[tid * 8 + j] = active ? src[j] : 0.f;
     __syncthreads();
     if (rs == 0 && ch8 < C8) {
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float acc = 0.f;
         for (int r = 0; r < rpb; ++r) acc += red[(r * cpb + cl) * 8 + j];
-        atomicAdd(&sum_dxhat[ch8 * 8 + j], acc);
+        sum_dpre[pass * plane + (long long)blockIdx.x * C + ch8 * 8 + j] = acc;
       }
     }
+    if (!NEED_XHAT) break;  // sx plane not needed in eval mode
   }
 }
 
-// vectorized bf16 BN backward apply (elementwise, 8 channels per lane)
+// coefficient precompute for the apply kernel: the training BN backward
+//   dx = gamma*invstd*(dpre - sum_dpre/M - xhat*sum_dxhat/M)
+// folds to dx = P[c]*dpre + Q[c]*x + R[c]; one thread per channel.
+__global__ void bn_bwd_coeffs_kernel(
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ sum_dpre,
+    const float* __restrict__ sum_dxhat, float* __restrict__ pqr, int C,
+    float invM) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float is = invstd[c];
+  float A = gamma[c] * is;
+  float P = A, Q = 0.f, R = 0.f;
+  if (sum_dpre) {
+    Q = -A * is * sum_dxhat[c] * invM;
+    R = -A * sum_dpre[c] * invM - Q * mean[c];
+  }
+  pqr[c] = P;
+  pqr[C + c] = Q;
+  pqr[2 * C + c] = R;
+}
+
+// vectorized bf16 BN backward apply: dx = P[c]*dpre + Q[c]*x + R[c]
 __global__ void bn_act_bwd_apply_bf16v8(
     const ushort8* __restrict__ dpre, const ushort8* __restrict__ x,
-    ushort8* __restrict__ dx, const float* __restrict__ mean,
-    const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ sum_dpre, const float* __restrict__ sum_dxhat,
-    long long total8, int C8, float invM) {
+    ushort8* __restrict__ dx, const float* __restrict__ pqr,
+    long long total8, int C8, int training) {
+  const int C = C8 * 8;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total8;
        i += (long long)gridDim.x * blockDim.x) {
     const int c8 = (int)(i % C8) * 8;
     ushort8 g8 = dpre[i];
     ushort8 x8;
-    if (sum_dpre) x8 = x[i];
+    if (training) x8 = x[i];
     ushort8 out;
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int c = c8 + j;
-      float g = us2f(g8[j]);
-      const float is = invstd[c];
-      if (sum_dpre) {
-        float xhat = (us2f(x8[j]) - mean[c]) * is;
-        g = g - sum_dpre[c] * invM - xhat * sum_dxhat[c] * invM;
-      }
-      out[j] = f2us(gamma[c] * is * g);
+      float v = pqr[c] * us2f(g8[j]);
+      if (training) v += pqr[C + c] * us2f(x8[j]) + pqr[2 * C + c];
+      out[j] = f2us(v);
     }
     dx[i] = out;
   }
@@ -314,13 +336,22 @@ static inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
+static inline bool use_v8() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("IBP_BN_V8");
+    v = (e == nullptr || e[0] != '0') ? 1 : 0;
+  }
+  return v == 1;
+}
+
 // pick the (channel-lanes, row-groups, row-blocks) geometry for the v8 kernels
 static inline void v8_geometry(long long M, int C8, int& cpb, int& rpb,
                                int& rows) {
   cpb = std::min(C8, 256);
   rpb = 256 / cpb;
-  // >=4 rows of work per thread, capped so the atomic traffic stays small
-  rows = (int)std::min<long long>((M + (long long)rpb * 4 - 1) / (rpb * 4), 1024);
+  // >=8 rows of work per thread; rows bounds the stage-2 column reduction
+  rows = (int)std::min<long long>((M + (long long)rpb * 8 - 1) / (rpb * 8), 512);
   rows = std::max(rows, 1);
 }
 
@@ -328,23 +359,28 @@ std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
   TORCH_CHECK(x_mc.is_cuda() && dense_ok(x_mc));
   long long M = x_mc.numel() / C;
   auto opts = x_mc.options().dtype(torch::kFloat32);
+  dim3 block(256);
+  if (use_v8() && x_mc.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
+    int C8 = (int)C / 8, cpb, rpb, rows;
+    v8_geometry(M, C8, cpb, rpb, rows);
+    Tensor ws = torch::empty({2LL * rows * C}, opts);
+    Tensor both = torch::empty({2, C}, opts);
+    dim3 grid(rows, (C8 + cpb - 1) / cpb);
+    hipLaunchKernelGGL(ibp::bn_stats_bf16v8, grid, block, 0, cur_stream(),
+                       reinterpret_cast<const ibp::ushort8*>(x_mc.data_ptr()),
+                       ws.data_ptr<float>(), M, C8, cpb, rpb);
+    dim3 cgrid((2 * C + 255) / 256);
+    hipLaunchKernelGGL(ibp::colsum_kernel, cgrid, block, 0, cur_stream(),
+                       ws.data_ptr<float>(), both.data_ptr<float>(),
+                       rows, (int)C, 2);
+    return {both[0], both[1]};
+  }
   // one allocation + one async memset for both accumulators (a torch::zeros
   // pair costs two FillFunctor launches per conv layer — visible in rocprof)
   Tensor both = torch::empty({2, C}, opts);
   hipMemsetAsync(both.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
   Tensor sums = both[0];
   Tensor sumsq = both[1];
-  dim3 block(256);
-  if (x_mc.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
-    int C8 = (int)C / 8, cpb, rpb, rows;
-    v8_geometry(M, C8, cpb, rpb, rows);
-    dim3 grid(rows, (C8 + cpb - 1) / cpb);
-    hipLaunchKernelGGL(ibp::bn_stats_bf16v8, grid, block, 0, cur_stream(),
-                       reinterpret_cast<const ibp::ushort8*>(x_mc.data_ptr()),
-                       sums.data_ptr<float>(), sumsq.data_ptr<float>(),
-                       M, C8, cpb, rpb);
-    return {sums, sumsq};
-  }
   int rows = (int)std::min<long long>((M + 63) / 64, 1024);
   dim3 grid(rows, (C + 255) / 256);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
@@ -403,31 +439,42 @@ std::vector<Tensor> bn_act_bwd(const Tensor& dy, const Tensor& y, const Tensor& 
   long long M = dy.numel() / C;
   auto fopts = dy.options().dtype(torch::kFloat32);
   Tensor dpre = torch::empty_like(dy);
-  Tensor both = torch::empty({2, C}, fopts);
-  hipMemsetAsync(both.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
-  Tensor sum_dpre = both[0];
-  Tensor sum_dxhat = both[1];
   dim3 block(256);
-  if (dy.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
+  if (use_v8() && dy.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
     int C8 = (int)C / 8, cpb, rpb, rows;
     v8_geometry(M, C8, cpb, rpb, rows);
+    Tensor ws = torch::empty({2LL * rows * C}, fopts);
+    Tensor both = torch::empty({2, C}, fopts);
+    if (!need_xhat) {
+      // only the sum_dpre plane is produced; second plane must read as zero
+      hipMemsetAsync(ws.data_ptr<float>() + (long long)rows * C, 0,
+                     (size_t)rows * C * sizeof(float), cur_stream());
+    }
     dim3 gridv(rows, (C8 + cpb - 1) / cpb);
-    auto launch = [&](auto need_xhat) {
-      hipLaunchKernelGGL((ibp::bn_act_bwd_reduce_bf16v8<decltype(need_xhat)::value>),
+    auto launch = [&](auto need) {
+      hipLaunchKernelGGL((ibp::bn_act_bwd_reduce_bf16v8<decltype(need)::value>),
                          gridv, block, 0, cur_stream(),
                          reinterpret_cast<const ibp::ushort8*>(dy.data_ptr()),
                          reinterpret_cast<const ibp::ushort8*>(y.data_ptr()),
                          reinterpret_cast<const ibp::ushort8*>(x.data_ptr()),
                          reinterpret_cast<ibp::ushort8*>(dpre.data_ptr()),
-                         need_xhat.value ? mean->data_ptr<float>() : nullptr,
-                         need_xhat.value ? invstd->data_ptr<float>() : nullptr,
-                         sum_dpre.data_ptr<float>(), sum_dxhat.data_ptr<float>(),
+                         need.value ? mean->data_ptr<float>() : nullptr,
+                         need.value ? invstd->data_ptr<float>() : nullptr,
+                         ws.data_ptr<float>(),
                          M, C8, cpb, rpb, (float)slope, act ? 1 : 0);
     };
     if (need_xhat) launch(std::true_type{});
     else launch(std::false_type{});
-    return {dpre, sum_dpre, sum_dxhat};
+    dim3 cgrid((2 * C + 255) / 256);
+    hipLaunchKernelGGL(ibp::colsum_kernel, cgrid, block, 0, cur_stream(),
+                       ws.data_ptr<float>(), both.data_ptr<float>(),
+                       rows, (int)C, 2);
+    return {dpre, both[0], both[1]};
   }
+  Tensor both = torch::empty({2, C}, fopts);
+  hipMemsetAsync(both.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
+  Tensor sum_dpre = both[0];
+  Tensor sum_dxhat = both[1];
   int rows = (int)std::min<long long>((M + 63) / 64, 1024);
   dim3 grid(rows, (C + 255) / 256);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
@@ -465,19 +512,25 @@ Tensor bn_act_bwd_apply(const Tensor& dpre, const Tensor& x, const Tensor& mean,
   long long M = dpre.numel() / C;
   Tensor dx = torch::empty_like(dpre);
   dim3 block(256);
-  if (dpre.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
+  if (use_v8() && dpre.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0) {
     long long total8 = dpre.numel() / 8;
+    Tensor pqr = torch::empty({3 * C}, dpre.options().dtype(torch::kFloat32));
+    bool training = sum_dpre.has_value();
+    dim3 pgrid(((int)C + 255) / 256);
+    hipLaunchKernelGGL(ibp::bn_bwd_coeffs_kernel, pgrid, block, 0, cur_stream(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(),
+                       training ? sum_dpre->data_ptr<float>() : nullptr,
+                       training ? sum_dxhat->data_ptr<float>() : nullptr,
+                       pqr.data_ptr<float>(), (int)C, 1.0f / (float)M);
     dim3 gridv(ibp::grid_1d(total8, 256, 8192));
     hipLaunchKernelGGL(ibp::bn_act_bwd_apply_bf16v8, gridv, block, 0,
                        cur_stream(),
                        reinterpret_cast<const ibp::ushort8*>(dpre.data_ptr()),
                        reinterpret_cast<const ibp::ushort8*>(x.data_ptr()),
                        reinterpret_cast<ibp::ushort8*>(dx.data_ptr()),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       gamma.data_ptr<float>(),
-                       sum_dpre.has_value() ? sum_dpre->data_ptr<float>() : nullptr,
-                       sum_dxhat.has_value() ? sum_dxhat->data_ptr<float>() : nullptr,
-                       total8, (int)C / 8, 1.0f / (float)M);
+                       pqr.data_ptr<float>(), total8, (int)C / 8,
+                       training ? 1 : 0);
     return dx;
   }
   dim3 grid(ibp::grid_1d(dpre.numel(), 256, 8192));
