@@ -91,19 +91,38 @@ __global__ void bn_stats_bf16v8(const ushort8* __restrict__ x,
   }
 }
 
-// stage 2: out[p][c] = sum_r ws[p][r][c]; one thread per (plane, channel)
+// stage 2: out[p][c] = sum_r ws[p][r][c]. 64 channel-lanes x 4 row-lanes per
+// block, 8 independent accumulators per thread -> 32 loads in flight per
+// lane-group (a single serial column walk costs rows x ~0.3us of latency).
 __global__ void colsum_kernel(const float* __restrict__ ws,
                               float* __restrict__ out, int rows, int C,
                               int planes) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= C * planes) return;
-  int p = i / C;
-  int c = i - p * C;
-  const float* src = ws + (long long)p * rows * C + c;
-  float acc = 0.f;
-  for (int r = 0; r < rows; ++r) acc += src[(long long)r * C];
-  out[i] = acc;
+  __shared__ float red[256];
+  const int total = C * planes;
+  const int cl = threadIdx.x & 63;
+  const int rl = threadIdx.x >> 6;  // 0..3
+  const int i = blockIdx.x * 64 + cl;
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  if (i < total) {
+    const int p = i / C;
+    const int c = i - p * C;
+    const float* src = ws + (long long)p * rows * C + c;
+    for (int r = rl; r < rows; r += 32) {
+      #pragma unroll
+      for (int u = 0; u < 8; ++u)
+        if (r + u * 4 < rows) acc[u] += src[(long long)(r + u * 4) * C];
+    }
+  }
+  float a = 0.f;
+  #pragma unroll
+  for (int u = 0; u < 8; ++u) a += acc[u];
+  red[threadIdx.x] = a;
+  __syncthreads();
+  if (rl == 0 && i < total)
+    out[i] = red[cl] + red[64 + cl] + red[128 + cl] + red[192 + cl];
 }
+
+typedef float float4w __attribute__((ext_vector_type(4)));
 
 __global__ void bn_act_fwd_bf16v8(const unsigned short* __restrict__ x,
                                   const unsigned short* __restrict__ res,
@@ -114,22 +133,27 @@ __global__ void bn_act_fwd_bf16v8(const unsigned short* __restrict__ x,
   const ushort8* xv = reinterpret_cast<const ushort8*>(x);
   const ushort8* rv = reinterpret_cast<const ushort8*>(res);
   ushort8* yv = reinterpret_cast<ushort8*>(y);
+  const float4w* sc4 = reinterpret_cast<const float4w*>(scale);
+  const float4w* sh4 = reinterpret_cast<const float4w*>(shift);
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total8;
        i += (long long)gridDim.x * blockDim.x) {
-    int cbase = (int)((i % C8) * 8);
+    const int q8 = (int)(i % C8) * 2;
+    float4w sc[2] = {sc4[q8], sc4[q8 + 1]};
+    float4w sh[2] = {sh4[q8], sh4[q8 + 1]};
     ushort8 xi = xv[i];
     ushort8 out;
     if (res != nullptr) {
       ushort8 ri = rv[i];
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float v = us2f(xi[j]) * scale[cbase + j] + shift[cbase + j] + us2f(ri[j]);
+        float v = us2f(xi[j]) * sc[j >> 2][j & 3] + sh[j >> 2][j & 3]
+                  + us2f(ri[j]);
         out[j] = f2us(act ? leaky(v, slope) : v);
       }
     } else {
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float v = us2f(xi[j]) * scale[cbase + j] + shift[cbase + j];
+        float v = us2f(xi[j]) * sc[j >> 2][j & 3] + sh[j >> 2][j & 3];
         out[j] = f2us(act ? leaky(v, slope) : v);
       }
     }
@@ -226,24 +250,37 @@ __global__ void bn_bwd_coeffs_kernel(
   pqr[2 * C + c] = R;
 }
 
-// vectorized bf16 BN backward apply: dx = P[c]*dpre + Q[c]*x + R[c]
+typedef float float4v __attribute__((ext_vector_type(4)));
+
+// vectorized bf16 BN backward apply: dx = P[c]*dpre + Q[c]*x + R[c].
+// Per-channel coefficients are fetched as float4 pairs — 6 dwordx4 instead of
+// 24 scalar dword loads per iteration (scalar form was vmem-issue bound).
 __global__ void bn_act_bwd_apply_bf16v8(
     const ushort8* __restrict__ dpre, const ushort8* __restrict__ x,
     ushort8* __restrict__ dx, const float* __restrict__ pqr,
     long long total8, int C8, int training) {
   const int C = C8 * 8;
+  const float4v* pqrv = reinterpret_cast<const float4v*>(pqr);
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total8;
        i += (long long)gridDim.x * blockDim.x) {
-    const int c8 = (int)(i % C8) * 8;
+    const int q8 = (int)(i % C8) * 2;  // float4-group index of channel c8
+    float4v P[2], Q[2], R[2];
+    P[0] = pqrv[q8];
+    P[1] = pqrv[q8 + 1];
+    if (training) {
+      Q[0] = pqrv[C / 4 + q8];
+      Q[1] = pqrv[C / 4 + q8 + 1];
+      R[0] = pqrv[C / 2 + q8];
+      R[1] = pqrv[C / 2 + q8 + 1];
+    }
     ushort8 g8 = dpre[i];
     ushort8 x8;
     if (training) x8 = x[i];
     ushort8 out;
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int c = c8 + j;
-      float v = pqr[c] * us2f(g8[j]);
-      if (training) v += pqr[C + c] * us2f(x8[j]) + pqr[2 * C + c];
+      float v = P[j >> 2][j & 3] * us2f(g8[j]);
+      if (training) v += Q[j >> 2][j & 3] * us2f(x8[j]) + R[j >> 2][j & 3];
       out[j] = f2us(v);
     }
     dx[i] = out;
