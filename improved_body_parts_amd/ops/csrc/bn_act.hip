@@ -406,7 +406,7 @@ std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
     hipLaunchKernelGGL(ibp::bn_stats_bf16v8, grid, block, 0, cur_stream(),
                        reinterpret_cast<const ibp::ushort8*>(x_mc.data_ptr()),
                        ws.data_ptr<float>(), M, C8, cpb, rpb);
-    dim3 cgrid((2 * C + 255) / 256);
+    dim3 cgrid((2 * (int)C + 63) / 64);  // colsum covers 64 channels/block
     hipLaunchKernelGGL(ibp::colsum_kernel, cgrid, block, 0, cur_stream(),
                        ws.data_ptr<float>(), both.data_ptr<float>(),
                        rows, (int)C, 2);
@@ -502,7 +502,7 @@ std::vector<Tensor> bn_act_bwd(const Tensor& dy, const Tensor& y, const Tensor& 
     };
     if (need_xhat) launch(std::true_type{});
     else launch(std::false_type{});
-    dim3 cgrid((2 * C + 255) / 256);
+    dim3 cgrid((2 * (int)C + 63) / 64);  // colsum covers 64 channels/block
     hipLaunchKernelGGL(ibp::colsum_kernel, cgrid, block, 0, cur_stream(),
                        ws.data_ptr<float>(), both.data_ptr<float>(),
                        rows, (int)C, 2);
