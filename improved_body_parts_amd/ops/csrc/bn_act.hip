@@ -61,8 +61,24 @@ __global__ void bn_stats_bf16v8(const ushort8* __restrict__ x,
   float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float q[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   if (active) {
-    for (long long m = (long long)blockIdx.x * rpb + rs; m < M;
-         m += (long long)gridDim.x * rpb) {
+    const long long step = (long long)gridDim.x * rpb;
+    long long m = (long long)blockIdx.x * rpb + rs;
+    // 4 rows in flight per thread: a single serial load chain leaves the
+    // kernel latency-bound at ~2 TB/s (measured)
+    for (; m + 3 * step < M; m += 4 * step) {
+      ushort8 x0 = x[m * C8 + ch8];
+      ushort8 x1 = x[(m + step) * C8 + ch8];
+      ushort8 x2 = x[(m + 2 * step) * C8 + ch8];
+      ushort8 x3 = x[(m + 3 * step) * C8 + ch8];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v0 = us2f(x0[j]), v1 = us2f(x1[j]);
+        float v2 = us2f(x2[j]), v3 = us2f(x3[j]);
+        s[j] += (v0 + v1) + (v2 + v3);
+        q[j] += (v0 * v0 + v1 * v1) + (v2 * v2 + v3 * v3);
+      }
+    }
+    for (; m < M; m += step) {
       ushort8 xv = x[m * C8 + ch8];
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -187,8 +203,36 @@ __global__ void bn_act_bwd_reduce_bf16v8(
     }
   }
   if (active) {
-    for (long long m = (long long)blockIdx.x * rpb + rs; m < M;
-         m += (long long)gridDim.x * rpb) {
+    const long long step = (long long)gridDim.x * rpb;
+    long long m = (long long)blockIdx.x * rpb + rs;
+    // 2 rows in flight (3 input streams each — 6 loads outstanding)
+    for (; m + step < M; m += 2 * step) {
+      const long long i0 = m * C8 + ch8;
+      const long long i1 = (m + step) * C8 + ch8;
+      ushort8 gd0 = dy[i0], gd1 = dy[i1];
+      ushort8 yy0, yy1;
+      if (act) { yy0 = y[i0]; yy1 = y[i1]; }
+      ushort8 xx0, xx1;
+      if (NEED_XHAT) { xx0 = x[i0]; xx1 = x[i1]; }
+      ushort8 o0, o1;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g0 = us2f(gd0[j]), g1 = us2f(gd1[j]);
+        if (act) {
+          g0 = us2f(yy0[j]) > 0.f ? g0 : g0 * slope;
+          g1 = us2f(yy1[j]) > 0.f ? g1 : g1 * slope;
+        }
+        o0[j] = f2us(g0);
+        o1[j] = f2us(g1);
+        s[j] += g0 + g1;
+        if (NEED_XHAT)
+          sx[j] += (g0 * (us2f(xx0[j]) - mu[j]) + g1 * (us2f(xx1[j]) - mu[j]))
+                   * is[j];
+      }
+      dpre_out[i0] = o0;
+      dpre_out[i1] = o1;
+    }
+    for (; m < M; m += step) {
       const long long i = m * C8 + ch8;
       ushort8 gd = dy[i];
       ushort8 yy;
