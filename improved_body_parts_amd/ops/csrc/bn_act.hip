@@ -110,23 +110,23 @@ __global__ void bn_stats_bf16v8(const ushort8* __restrict__ x,
 // stage 2: out[p][c] = sum_r ws[p][r][c]. 64 channel-lanes x 4 row-lanes per
 // block, 8 independent accumulators per thread -> 32 loads in flight per
 // lane-group (a single serial column walk costs rows x ~0.3us of latency).
-__global__ void colsum_kernel(const float* __restrict__ ws,
-                              float* __restrict__ out, int rows, int C,
-                              int planes) {
-  __shared__ float red[256];
+__global__ __launch_bounds__(1024) void colsum_kernel(
+    const float* __restrict__ ws, float* __restrict__ out, int rows, int C,
+    int planes) {
+  __shared__ float red[1024];
   const int total = C * planes;
   const int cl = threadIdx.x & 63;
-  const int rl = threadIdx.x >> 6;  // 0..3
+  const int rl = threadIdx.x >> 6;  // 0..15
   const int i = blockIdx.x * 64 + cl;
   float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   if (i < total) {
     const int p = i / C;
     const int c = i - p * C;
     const float* src = ws + (long long)p * rows * C + c;
-    for (int r = rl; r < rows; r += 32) {
+    for (int r = rl; r < rows; r += 128) {
       #pragma unroll
       for (int u = 0; u < 8; ++u)
-        if (r + u * 4 < rows) acc[u] += src[(long long)(r + u * 4) * C];
+        if (r + u * 16 < rows) acc[u] += src[(long long)(r + u * 16) * C];
     }
   }
   float a = 0.f;
@@ -134,8 +134,11 @@ __global__ void colsum_kernel(const float* __restrict__ ws,
   for (int u = 0; u < 8; ++u) a += acc[u];
   red[threadIdx.x] = a;
   __syncthreads();
-  if (rl == 0 && i < total)
-    out[i] = red[cl] + red[64 + cl] + red[128 + cl] + red[192 + cl];
+  if (rl == 0 && i < total) {
+    #pragma unroll
+    for (int g = 1; g < 16; ++g) a += red[g * 64 + cl];
+    out[i] = a;
+  }
 }
 
 typedef float float4w __attribute__((ext_vector_type(4)));
@@ -451,7 +454,7 @@ std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
                        reinterpret_cast<const ibp::ushort8*>(x_mc.data_ptr()),
                        ws.data_ptr<float>(), M, C8, cpb, rpb);
     dim3 cgrid((2 * (int)C + 63) / 64);  // colsum covers 64 channels/block
-    hipLaunchKernelGGL(ibp::colsum_kernel, cgrid, block, 0, cur_stream(),
+    hipLaunchKernelGGL(ibp::colsum_kernel, cgrid, dim3(1024), 0, cur_stream(),
                        ws.data_ptr<float>(), both.data_ptr<float>(),
                        rows, (int)C, 2);
     return {both[0], both[1]};
@@ -547,7 +550,7 @@ std::vector<Tensor> bn_act_bwd(const Tensor& dy, const Tensor& y, const Tensor& 
     if (need_xhat) launch(std::true_type{});
     else launch(std::false_type{});
     dim3 cgrid((2 * (int)C + 63) / 64);  // colsum covers 64 channels/block
-    hipLaunchKernelGGL(ibp::colsum_kernel, cgrid, block, 0, cur_stream(),
+    hipLaunchKernelGGL(ibp::colsum_kernel, cgrid, dim3(1024), 0, cur_stream(),
                        ws.data_ptr<float>(), both.data_ptr<float>(),
                        rows, (int)C, 2);
     return {dpre, both[0], both[1]};
