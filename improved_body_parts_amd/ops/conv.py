@@ -101,22 +101,20 @@ class ConvBnActFn(torch.autograd.Function):
 
         if gamma is not None:
             if training:
-                sums, sumsq = ext.bn_stats(y_conv, C)
-                M = y_conv.numel() // C
-                mean = sums / M
-                var = (sumsq / M - mean * mean).clamp_(min=0.0)
-                with torch.no_grad():
-                    mom = bn_mod.momentum if bn_mod.momentum is not None else 0.1
-                    unbiased = var * (M / max(M - 1, 1))
-                    bn_mod.running_mean.mul_(1 - mom).add_(mean, alpha=mom)
-                    bn_mod.running_var.mul_(1 - mom).add_(unbiased, alpha=mom)
-                    bn_mod.num_batches_tracked += 1
+                # single fused kernel chain: stats + colsum + per-channel
+                # epilogue + running-stat update (the Python mean/var/rsqrt
+                # chain was ~6 tiny launches per conv layer)
+                mom = bn_mod.momentum if bn_mod.momentum is not None else 0.1
+                mean, invstd, scale, shift = ext.bn_stats_finalize(
+                    y_conv, C, gamma.float(), beta.float(),
+                    bn_mod.running_mean, bn_mod.running_var,
+                    bn_mod.num_batches_tracked, mom, bn_mod.eps)
             else:
                 mean = bn_mod.running_mean.float()
                 var = bn_mod.running_var.float()
-            invstd = torch.rsqrt(var + bn_mod.eps)
-            scale = (gamma.float() * invstd)
-            shift = (beta.float() - mean * scale)
+                invstd = torch.rsqrt(var + bn_mod.eps)
+                scale = (gamma.float() * invstd)
+                shift = (beta.float() - mean * scale)
         elif act or residual is not None:
             mean = invstd = None
             scale = torch.ones(C, device=x.device, dtype=torch.float32)

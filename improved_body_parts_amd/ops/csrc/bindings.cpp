@@ -5,6 +5,11 @@ using torch::Tensor;
 
 // bn_act.hip
 std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C);
+std::vector<Tensor> bn_stats_finalize(
+    const Tensor& x_mc, int64_t C, const Tensor& gamma, const Tensor& beta,
+    const c10::optional<Tensor>& running_mean,
+    const c10::optional<Tensor>& running_var,
+    const c10::optional<Tensor>& num_batches, double momentum, double eps);
 Tensor bn_act_fwd(const Tensor& x, const Tensor& scale, const Tensor& shift,
                   const c10::optional<Tensor>& residual, double slope, bool act);
 std::vector<Tensor> bn_act_bwd(const Tensor& dy, const Tensor& y, const Tensor& x,
@@ -66,6 +71,8 @@ Tensor limb_scores(const Tensor& paf, const Tensor& peaks, const Tensor& cand_id
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_stats", &bn_stats, "per-channel sum/sumsq of an [M][C] view");
+  m.def("bn_stats_finalize", &bn_stats_finalize,
+        "stats + mean/invstd/scale/shift epilogue (+ running-stat update)");
   m.def("bn_act_fwd", &bn_act_fwd, "fused scale/shift (+res) (+leaky)");
   m.def("bn_act_bwd", &bn_act_bwd, "dpre + per-channel reductions");
   m.def("bn_act_bwd_apply", &bn_act_bwd_apply, "BN backward input grad");

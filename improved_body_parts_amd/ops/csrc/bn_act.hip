@@ -141,6 +141,35 @@ __global__ __launch_bounds__(1024) void colsum_kernel(
   }
 }
 
+// fold the per-channel BN statistics epilogue into ONE kernel: the Python
+// mean/var/rsqrt/scale/shift chain was ~6 tiny fp32 launches per conv layer
+// (~8% of the training step, rocprof). Updates running stats in place.
+__global__ void bn_finalize_kernel(
+    const float* __restrict__ sums, const float* __restrict__ sumsq,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    long long* __restrict__ num_batches, float* __restrict__ out /*[4][C]*/,
+    long long M, int C, float momentum, float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float invM = 1.0f / (float)M;
+  float mean = sums[c] * invM;
+  float var = fmaxf(sumsq[c] * invM - mean * mean, 0.f);
+  float invstd = rsqrtf(var + eps);
+  if (running_mean) {
+    float unbiased = var * ((float)M / (float)(M > 1 ? M - 1 : 1));
+    running_mean[c] = running_mean[c] * (1.f - momentum) + mean * momentum;
+    running_var[c] = running_var[c] * (1.f - momentum) + unbiased * momentum;
+  }
+  float scale = gamma[c] * invstd;
+  float shift = beta[c] - mean * scale;
+  out[c] = mean;
+  out[C + c] = invstd;
+  out[2 * C + c] = scale;
+  out[3 * C + c] = shift;
+  if (c == 0 && num_batches) *num_batches += 1;
+}
+
 typedef float float4w __attribute__((ext_vector_type(4)));
 
 __global__ void bn_act_fwd_bf16v8(const unsigned short* __restrict__ x,
@@ -475,6 +504,32 @@ std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
                        sums.data_ptr<float>(), sumsq.data_ptr<float>(), M, (int)C);
   });
   return {sums, sumsq};
+}
+
+// stats + column-sum + per-channel epilogue in one call; returns
+// [mean, invstd, scale, shift] (views of one [4][C] tensor). Training path:
+// also updates running_mean/var (+ num_batches_tracked) in place.
+std::vector<Tensor> bn_stats_finalize(
+    const Tensor& x_mc, int64_t C, const Tensor& gamma, const Tensor& beta,
+    const c10::optional<Tensor>& running_mean,
+    const c10::optional<Tensor>& running_var,
+    const c10::optional<Tensor>& num_batches, double momentum, double eps) {
+  TORCH_CHECK(gamma.scalar_type() == at::ScalarType::Float &&
+              beta.scalar_type() == at::ScalarType::Float,
+              "bn_stats_finalize expects fp32 BN affine parameters");
+  auto sv = bn_stats(x_mc, C);
+  long long M = x_mc.numel() / C;
+  Tensor out = torch::empty({4, C}, x_mc.options().dtype(torch::kFloat32));
+  dim3 block(256), grid(((int)C + 255) / 256);
+  hipLaunchKernelGGL(ibp::bn_finalize_kernel, grid, block, 0, cur_stream(),
+                     sv[0].data_ptr<float>(), sv[1].data_ptr<float>(),
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                     running_mean ? running_mean->data_ptr<float>() : nullptr,
+                     running_var ? running_var->data_ptr<float>() : nullptr,
+                     num_batches ? num_batches->data_ptr<long long>() : nullptr,
+                     out.data_ptr<float>(), M, (int)C, (float)momentum,
+                     (float)eps);
+  return {out[0], out[1], out[2], out[3]};
 }
 
 Tensor bn_act_fwd(const Tensor& x, const Tensor& scale, const Tensor& shift,
