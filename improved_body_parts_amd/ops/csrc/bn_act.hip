@@ -526,7 +526,7 @@ std::vector<Tensor> bn_stats_finalize(
                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      running_mean ? running_mean->data_ptr<float>() : nullptr,
                      running_var ? running_var->data_ptr<float>() : nullptr,
-                     num_batches ? num_batches->data_ptr<long long>() : nullptr,
+                     num_batches ? reinterpret_cast<long long*>(num_batches->data_ptr<int64_t>()) : nullptr,
                      out.data_ptr<float>(), M, (int)C, (float)momentum,
                      (float)eps);
   return {out[0], out[1], out[2], out[3]};
