@@ -49,7 +49,6 @@ struct ConvParams {
   int act;                   // leaky-relu on epilogue
   int splitk;                // K-dimension split factor (small-grid layers)
   float* ws;                 // fp32 workspace for split-K partial accumulation
-  int* cnt;                  // per-tile arrival counters (split-K reduction)
 };
 
 // LDS tile addressing: unpadded 128-B rows with a T2 XOR swizzle — the 16-lane
@@ -264,42 +263,25 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
       }
     }
   }
+}
 
-  // ---- in-launch split-K reduction (cdna_hip_programming.md §6 G16, counter
-  // form): each slice block releases its fp32 slab, takes a ticket; the block
-  // drawing splitk-1 acquires and reduces — saves the combine-kernel boundary
-  // (7.4 us x ~210 calls/step on the inference path, rocprof).
-  if (p.splitk > 1) {
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    if (tid == 0) {
-      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-      // ROCm 7.2 drops the fence's own vmcnt wait when its scoreboard thinks
-      // nothing is outstanding; restate it (guide §6 G16 pitfall 12)
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      int prev = __hip_atomic_fetch_add(&p.cnt[tile], 1, __ATOMIC_RELAXED,
-                                        __HIP_MEMORY_SCOPE_AGENT);
-      // broadcast "I am last" through the existing LDS array (no second
-      // __shared__ object)
-      lds_a[0][0] = (prev == p.splitk - 1) ? 1 : 0;
-    }
-    __syncthreads();
-    if (lds_a[0][0] == 0) return;
-    if (tid == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    __syncthreads();
-    const long long plane = p.M * (long long)p.Cout;
-    for (int idx = tid; idx < BM * BN; idx += 256) {
-      const long long m = m0 + idx / BN;
-      const int col = n0 + idx % BN;
-      if (m >= p.M || col >= p.Cout) continue;
-      const long long base = m * p.Cout + col;
-      float v = 0.f;
-      for (int s2 = 0; s2 < p.splitk; ++s2) v += p.ws[s2 * plane + base];
-      if (p.scale) v = v * p.scale[col] + p.shift[col];
-      if (p.res) v += us2f(p.res[base]);
-      if (p.act) v = leaky(v, 0.01f);
-      p.y[base] = f2us(v);
-    }
+// split-K combine: y = act(scale*ws + shift (+res)) -> bf16
+__global__ void splitk_combine_kernel(const float* __restrict__ ws,
+                                      unsigned short* __restrict__ y,
+                                      const float* __restrict__ scale,
+                                      const float* __restrict__ shift,
+                                      const unsigned short* __restrict__ res,
+                                      long long total, int C, int act,
+                                      int splitk) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float v = 0.f;
+    for (int s = 0; s < splitk; ++s) v += ws[(long long)s * total + i];
+    if (scale) v = v * scale[c] + shift[c];
+    if (res) v += us2f(res[i]);
+    if (act) v = leaky(v, 0.01f);
+    y[i] = f2us(v);
   }
 }
 
@@ -347,35 +329,34 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
   const int BN = (Cout > 64 && padded128 <= padded64) ? 128 : 64;
   const int ntiles = p.n_mtiles * (int)((Cout + BN - 1) / BN);
   const int nk_total = (p.K + ibp::BK - 1) / ibp::BK;
-  // split K on small grids so the 256-CU chip stays filled (~2 blocks/CU);
-  // normalise so EVERY slice has work (an empty slice would never arrive at
-  // the in-kernel reduction's ticket counter)
+  // split K on small grids so the 256-CU chip stays filled (~2 blocks/CU)
   int splitk = 1;
   if (ntiles < 384 && nk_total > 1) {
     splitk = std::min((int)nk_total, (384 + ntiles - 1) / ntiles);
-    int nk_chunk = ((int)nk_total + splitk - 1) / splitk;
-    splitk = ((int)nk_total + nk_chunk - 1) / nk_chunk;
   }
   p.splitk = splitk;
-  Tensor ws, cnt;
+  Tensor ws;
   if (splitk > 1) {
     // per-split slices written with plain stores -> empty() is safe (every
     // in-range element is covered by every split's epilogue)
     ws = torch::empty({(long long)splitk * p.M * Cout},
                       x.options().dtype(torch::kFloat32));
     p.ws = ws.data_ptr<float>();
-    cnt = torch::empty({ntiles}, x.options().dtype(torch::kInt32));
-    p.cnt = cnt.data_ptr<int>();
-    hipMemsetAsync(p.cnt, 0, ntiles * sizeof(int), stream);
   } else {
     p.ws = nullptr;
-    p.cnt = nullptr;
   }
   dim3 grid(ntiles * splitk), block(256);
   if (BN == 128) {
     hipLaunchKernelGGL(ibp::conv_mfma_kernel<128>, grid, block, 0, stream, p);
   } else {
     hipLaunchKernelGGL(ibp::conv_mfma_kernel<64>, grid, block, 0, stream, p);
+  }
+  if (splitk > 1) {
+    long long total = (long long)p.M * Cout;
+    dim3 cgrid(ibp::grid_1d(total, 256, 4096)), cblock(256);
+    hipLaunchKernelGGL(ibp::splitk_combine_kernel, cgrid, cblock, 0, stream,
+                       p.ws, p.y, p.scale, p.shift, p.res, total, (int)Cout,
+                       p.act, splitk);
   }
   return y;
 }
