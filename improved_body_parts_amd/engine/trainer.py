@@ -185,7 +185,7 @@ class Trainer:
         if self.reducer is not None:
             self.reducer.zero_grad()
         else:
-            self.optimizer.zero_grad(set_to_none=False)
+            self.optimizer.zero_grad(set_to_none=True)
         loss = self.model((images, mask_miss, heatmaps))
         lv = float(loss.detach())
         if lv > self.opt.loss_explosion_thre or lv != lv:
