@@ -99,16 +99,30 @@ class ConvBnActFn(torch.autograd.Function):
                                stride, padding, dilation)
         C = y_conv.shape[1]
 
+        sync_group, sync_world = _sync_info(bn_mod, training)
         if gamma is not None:
             if training:
-                # single fused kernel chain: stats + colsum + per-channel
-                # epilogue + running-stat update (the Python mean/var/rsqrt
-                # chain was ~6 tiny launches per conv layer)
                 mom = bn_mod.momentum if bn_mod.momentum is not None else 0.1
-                mean, invstd, scale, shift = ext.bn_stats_finalize(
-                    y_conv, C, gamma.float(), beta.float(),
-                    bn_mod.running_mean, bn_mod.running_var,
-                    bn_mod.num_batches_tracked, mom, bn_mod.eps)
+                if sync_world > 1:
+                    # SyncBN: all-reduce (sums, sumsq) so the fused path keeps
+                    # cross-rank statistics (module forward is bypassed here)
+                    import torch.distributed as dist
+                    sums, sumsq = ext.bn_stats(y_conv, C)
+                    flat = torch.cat([sums, sumsq])
+                    dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=sync_group)
+                    M = (y_conv.numel() // C) * sync_world
+                    mean, invstd, scale, shift = ext.bn_finalize(
+                        flat[:C], flat[C:], M, gamma.float(), beta.float(),
+                        bn_mod.running_mean, bn_mod.running_var,
+                        bn_mod.num_batches_tracked, mom, bn_mod.eps)
+                else:
+                    # single fused kernel chain: stats + colsum + per-channel
+                    # epilogue + running-stat update (the Python mean/var/rsqrt
+                    # chain was ~6 tiny launches per conv layer)
+                    mean, invstd, scale, shift = ext.bn_stats_finalize(
+                        y_conv, C, gamma.float(), beta.float(),
+                        bn_mod.running_mean, bn_mod.running_var,
+                        bn_mod.num_batches_tracked, mom, bn_mod.eps)
             else:
                 mean = bn_mod.running_mean.float()
                 var = bn_mod.running_var.float()
@@ -132,6 +146,7 @@ class ConvBnActFn(torch.autograd.Function):
         ctx.save_for_backward(x, weight, gamma, y_conv, y, mean, invstd)
         ctx.conf = (stride, padding, dilation, act, training,
                     residual is not None, bias is not None and gamma is None)
+        ctx.sync = (sync_group, sync_world)
         return y
 
     @staticmethod
@@ -156,6 +171,18 @@ class ConvBnActFn(torch.autograd.Function):
             dy, y, y_conv, mean if has_bn else None, invstd if has_bn else None,
             LEAKY_SLOPE, act, has_bn, C)
 
+        sync_group, sync_world = getattr(ctx, "sync", (None, 1))
+        if has_bn and training and sync_world > 1:
+            # SyncBN backward: dx needs GLOBAL sum_dpre/sum_dxhat over the
+            # global count; averaging (sum / world) makes the local-M division
+            # inside bn_act_bwd_apply equal the global-M division, and the
+            # per-rank dgamma/dbeta then DDP-average to the correct value.
+            import torch.distributed as dist
+            flat = torch.cat([sum_dpre, sum_dxhat])
+            dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=sync_group)
+            flat = flat / sync_world
+            sum_dpre, sum_dxhat = flat[:C], flat[C:]
+
         dres = dpre if has_res else None
         if has_bn:
             if training:
@@ -177,6 +204,19 @@ class ConvBnActFn(torch.autograd.Function):
             if ctx.needs_input_grad[1] else None
         return (dx, dw, dbias, dgamma, dbeta, dres,
                 None, None, None, None, None, None, None)
+
+
+def _sync_info(bn_mod, training):
+    """(process_group, world_size) when bn_mod is a SyncBatchNorm2d in a
+    multi-rank training run; (None, 1) otherwise."""
+    if not training or bn_mod is None or not hasattr(bn_mod, "process_group"):
+        return None, 1
+    import torch.distributed as dist
+    if not dist.is_initialized():
+        return None, 1
+    group = bn_mod.process_group
+    world = dist.get_world_size(group)
+    return group, world
 
 
 def conv_bn_act_hip(x, conv, bn, act: bool, residual=None, training: bool = False):

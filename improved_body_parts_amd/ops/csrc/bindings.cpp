@@ -5,6 +5,11 @@ using torch::Tensor;
 
 // bn_act.hip
 std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C);
+std::vector<Tensor> bn_finalize(
+    const Tensor& sums, const Tensor& sumsq, int64_t M, const Tensor& gamma,
+    const Tensor& beta, const c10::optional<Tensor>& running_mean,
+    const c10::optional<Tensor>& running_var,
+    const c10::optional<Tensor>& num_batches, double momentum, double eps);
 std::vector<Tensor> bn_stats_finalize(
     const Tensor& x_mc, int64_t C, const Tensor& gamma, const Tensor& beta,
     const c10::optional<Tensor>& running_mean,
@@ -71,6 +76,8 @@ Tensor limb_scores(const Tensor& paf, const Tensor& peaks, const Tensor& cand_id
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_stats", &bn_stats, "per-channel sum/sumsq of an [M][C] view");
+  m.def("bn_finalize", &bn_finalize,
+        "mean/invstd/scale/shift epilogue over given (possibly all-reduced) sums");
   m.def("bn_stats_finalize", &bn_stats_finalize,
         "stats + mean/invstd/scale/shift epilogue (+ running-stat update)");
   m.def("bn_act_fwd", &bn_act_fwd, "fused scale/shift (+res) (+leaky)");

@@ -506,6 +506,28 @@ std::vector<Tensor> bn_stats(const Tensor& x_mc, int64_t C) {
   return {sums, sumsq};
 }
 
+// per-channel epilogue over given sums (SyncBN path all-reduces the sums
+// between bn_stats and this); returns [mean, invstd, scale, shift].
+std::vector<Tensor> bn_finalize(
+    const Tensor& sums, const Tensor& sumsq, int64_t M, const Tensor& gamma,
+    const Tensor& beta, const c10::optional<Tensor>& running_mean,
+    const c10::optional<Tensor>& running_var,
+    const c10::optional<Tensor>& num_batches, double momentum, double eps) {
+  int64_t C = sums.numel();
+  Tensor out = torch::empty({4, C}, sums.options().dtype(torch::kFloat32));
+  dim3 block(256), grid(((int)C + 255) / 256);
+  hipLaunchKernelGGL(ibp::bn_finalize_kernel, grid, block, 0, cur_stream(),
+                     sums.data_ptr<float>(), sumsq.data_ptr<float>(),
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                     running_mean ? running_mean->data_ptr<float>() : nullptr,
+                     running_var ? running_var->data_ptr<float>() : nullptr,
+                     num_batches ? reinterpret_cast<long long*>(
+                         num_batches->data_ptr<int64_t>()) : nullptr,
+                     out.data_ptr<float>(), M, (int)C, (float)momentum,
+                     (float)eps);
+  return {out[0], out[1], out[2], out[3]};
+}
+
 // stats + column-sum + per-channel epilogue in one call; returns
 // [mean, invstd, scale, shift] (views of one [4][C] tensor). Training path:
 // also updates running_mean/var (+ num_batches_tracked) in place.
