@@ -48,6 +48,8 @@ def parse_args():
     p.add_argument("--nstack", type=int, default=4)
     p.add_argument("--input", type=int, default=512)
     p.add_argument("--no-bf16", action="store_true")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the inference forward in a hipGraph and replay")
     p.add_argument("--allow-eager", action="store_true",
                    help="permit eager fallback if the HIP extension is absent")
     return p.parse_args()
@@ -138,10 +140,27 @@ def main():
         img = torch.rand(batch, config.height, config.width, 3,
                          device=device, dtype=dtype)
 
-        @torch.no_grad()
-        def step(i):
-            out = model(img)
-            return out[-1][0]  # last stack, scale 0 (reference evaluate.py:126)
+        if args.graph and use_cuda:
+            # hipGraph capture: one replay per step, zero per-kernel launch
+            # overhead from the host
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s), torch.no_grad():
+                for _ in range(3):
+                    model(img)
+            torch.cuda.current_stream().wait_stream(s)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph), torch.no_grad():
+                static_out = model(img)
+
+            def step(i):
+                graph.replay()
+                return static_out[-1][0]
+        else:
+            @torch.no_grad()
+            def step(i):
+                out = model(img)
+                return out[-1][0]  # last stack, scale 0 (reference evaluate.py:126)
 
         metric_name = "fps_512_infer"
         vs_baseline = None  # filled below from the 38.5 FPS headline
