@@ -189,6 +189,23 @@ def DenseSkeletonConfig(width: int = 512, height: int = 512, stride: int = 4) ->
     return CanonicalConfig(width, height, stride, limbs=_LIMBS + _DENSE_EXTRA_LIMBS)
 
 
+# 24-limb slim skeleton of the 3-stage @384 variant (reference config2.py:69-83)
+_SLIM_LIMBS = [
+    ("neck", "nose"), ("neck", "Reye"), ("neck", "Leye"), ("neck", "Rear"),
+    ("neck", "Lear"), ("nose", "Reye"), ("nose", "Leye"), ("Reye", "Rear"),
+    ("Leye", "Lear"), ("neck", "Rsho"), ("Rsho", "Relb"), ("Relb", "Rwri"),
+    ("neck", "Lsho"), ("Lsho", "Lelb"), ("Lelb", "Lwri"), ("neck", "Rhip"),
+    ("Rhip", "Rkne"), ("Rkne", "Rank"), ("neck", "Lhip"), ("Lhip", "Lkne"),
+    ("Lkne", "Lank"), ("Rhip", "Lhip"), ("Rsho", "Rear"), ("Lsho", "Lear"),
+]
+
+
+def SlimSkeletonConfig(width: int = 384, height: int = 384, stride: int = 4) -> CanonicalConfig:
+    """24-limb 44-channel variant at 384^2 (capability of reference
+    config2.py's 3-stage configuration)."""
+    return CanonicalConfig(width, height, stride, limbs=_SLIM_LIMBS)
+
+
 class COCOSourceConfig:
     """COCO dataset joint order -> canonical order adapter (reference config/config.py:137-233)."""
 
@@ -238,6 +255,7 @@ Configs = {
     "Canonical384": lambda: CanonicalConfig(384, 384, 4),
     "Canonical768": lambda: CanonicalConfig(768, 768, 4),
     "DenseSkeleton": DenseSkeletonConfig,
+    "Slim384": SlimSkeletonConfig,
 }
 
 

@@ -87,3 +87,12 @@ def test_inference_params_reader(tmp_path):
     assert params["thre1"] == pytest.approx(0.25)
     assert params["scale_search"] == [0.5, 1.0, 1.5]
     assert model_params["boxsize"] == 320
+
+
+def test_slim384_variant():
+    """24-limb 44-channel @384 preset (reference config2.py capability)."""
+    c = GetConfig("Slim384")
+    assert c.paf_layers == 24
+    assert c.num_layers == 24 + 18 + 2
+    assert (c.width, c.height) == (384, 384)
+    assert len(c.flip_paf_ord) == 24
