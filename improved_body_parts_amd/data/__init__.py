@@ -5,12 +5,12 @@ Mirrors the capabilities of the reference's data/ + py_cocodata_server/ packages
 """
 from .heatmapper import Heatmapper, create_heatmaps_device, limb_gaussian
 from .transformer import Transformer, AugmentSelection
-from .synthetic import SyntheticPoseDataset, sample_people
+from .synthetic import SyntheticPoseDataset, DeviceGTSyntheticLoader, sample_people
 from .coco import MyDataset, RawDataIterator, build_coco_h5
 
 __all__ = [
     "Heatmapper", "create_heatmaps_device", "limb_gaussian",
     "Transformer", "AugmentSelection",
-    "SyntheticPoseDataset", "sample_people",
+    "SyntheticPoseDataset", "DeviceGTSyntheticLoader", "sample_people",
     "MyDataset", "RawDataIterator", "build_coco_h5",
 ]

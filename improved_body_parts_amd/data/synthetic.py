@@ -106,3 +106,70 @@ class SyntheticPoseDataset(Dataset):
         img, mask_miss, heatmaps, _ = self.generate(index)
         return (torch.from_numpy(img), torch.from_numpy(mask_miss),
                 torch.from_numpy(np.ascontiguousarray(heatmaps)))
+
+
+class DeviceGTSyntheticLoader:
+    """GPU-resident synthetic training stream (north-star: the full 512^2
+    pipeline stays on device).
+
+    Per batch: the tiny joint skeletons are sampled on the host (a few KB),
+    everything heavy happens on the GPU — random images via torch, ground-truth
+    heatmaps via the HIP batched generator (ops/csrc/heatmap_gt.hip, the device
+    twin of the numpy oracle the reference runs per DataLoader worker at
+    ~40 samples/s/process, reference README.md:35).
+
+    Iterable yielding ``(images (B,H,W,3), mask_miss (B,1,h,w),
+    heatmaps (B,C,h,w))`` on ``device`` in ``dtype``.
+    """
+
+    def __init__(self, config, batch_size: int, steps_per_epoch: int = 256,
+                 seed: int = 0, max_people: int = 4, device="cuda",
+                 dtype=torch.float32):
+        self.config = config
+        self.batch_size = batch_size
+        self.steps = steps_per_epoch
+        self.seed = seed
+        self.max_people = max_people
+        self.device = device
+        self.dtype = dtype
+        self._epoch = 0
+
+    def set_epoch(self, epoch: int):
+        self._epoch = epoch
+
+    def __len__(self):
+        return self.steps
+
+    def __iter__(self):
+        from .heatmapper import create_heatmaps_device
+        cfg = self.config
+        h, w = cfg.mask_shape
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(self.seed * 7_777_777 + self._epoch)
+        rng = np.random.default_rng(self.seed * 1_000_003 + self._epoch)
+        for _ in range(self.steps):
+            joints = np.full((self.batch_size, self.max_people,
+                              cfg.num_parts, 3), 2.0, dtype=np.float32)
+            masks = np.zeros((self.batch_size, h, w), dtype=np.float32)
+            for b in range(self.batch_size):
+                people = sample_people(rng, cfg.width, cfg.height,
+                                       self.max_people)
+                joints[b, :len(people)] = people
+                for p in people:
+                    marked = p[:, 2] < 2
+                    if not marked.any():
+                        continue
+                    xs = p[marked, 0] / cfg.stride
+                    ys = p[marked, 1] / cfg.stride
+                    x0, x1 = int(max(xs.min() - 2, 0)), int(min(xs.max() + 2, w))
+                    y0, y1 = int(max(ys.min() - 2, 0)), int(min(ys.max() + 2, h))
+                    masks[b, y0:y1, x0:x1] = 1.0
+            images = torch.rand(self.batch_size, cfg.height, cfg.width, 3,
+                                generator=gen, device=self.device,
+                                dtype=torch.float32)
+            heatmaps = create_heatmaps_device(joints, masks, cfg,
+                                              device=self.device)
+            mask_miss = torch.ones(self.batch_size, 1, h, w,
+                                   device=self.device)
+            yield (images.to(self.dtype), mask_miss.to(self.dtype),
+                   heatmaps.to(self.dtype))

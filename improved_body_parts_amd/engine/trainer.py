@@ -82,7 +82,7 @@ class Trainer:
     def __init__(self, opt, config, train_dataset, val_dataset=None, *,
                  rank=0, local_rank=0, world_size=1, use_bn=True, sync_bn=None,
                  num_workers=2, checkpoint_dir="checkpoints", log_file=None,
-                 device=None):
+                 device=None, device_synth_steps=0):
         self.opt = opt
         self.config = config
         self.rank = rank
@@ -114,13 +114,24 @@ class Trainer:
                                   weight_decay=opt.weight_decay)
         self.reducer = (GradReducer(self.model) if world_size > 1 else None)
 
-        self.train_sampler = (DistributedSampler(train_dataset)
-                              if world_size > 1 else None)
-        self.train_loader = DataLoader(
-            train_dataset, batch_size=opt.batch_size,
-            shuffle=self.train_sampler is None, sampler=self.train_sampler,
-            num_workers=num_workers, pin_memory=self.device.type == "cuda",
-            drop_last=True)
+        if device_synth_steps and self.device.type == "cuda":
+            # GPU-resident data: GT heatmaps from the HIP batched generator,
+            # different stream per rank (north-star: 512^2 pipeline on device)
+            from ..data import DeviceGTSyntheticLoader
+            self.train_sampler = None
+            self.train_loader = DeviceGTSyntheticLoader(
+                train_dataset if isinstance(train_dataset, int) else config,
+                batch_size=opt.batch_size, steps_per_epoch=device_synth_steps,
+                seed=rank + 1, device=self.device,
+                dtype=torch.bfloat16 if self.bf16 else torch.float32)
+        else:
+            self.train_sampler = (DistributedSampler(train_dataset)
+                                  if world_size > 1 else None)
+            self.train_loader = DataLoader(
+                train_dataset, batch_size=opt.batch_size,
+                shuffle=self.train_sampler is None, sampler=self.train_sampler,
+                num_workers=num_workers, pin_memory=self.device.type == "cuda",
+                drop_last=True)
         self.val_loader = (DataLoader(val_dataset, batch_size=opt.batch_size,
                                       shuffle=False, num_workers=num_workers,
                                       pin_memory=self.device.type == "cuda")
@@ -151,6 +162,8 @@ class Trainer:
         self.model.train()
         if self.train_sampler is not None:
             self.train_sampler.set_epoch(epoch)
+        elif hasattr(self.train_loader, "set_epoch"):
+            self.train_loader.set_epoch(epoch)  # device-GT stream
         meter = AverageMeter()
         iters_per_epoch = len(self.train_loader)
         t0 = time.time()
@@ -255,6 +268,8 @@ class SWATrainer(Trainer):
                 m.eval()
         if self.train_sampler is not None:
             self.train_sampler.set_epoch(epoch)
+        elif hasattr(self.train_loader, "set_epoch"):
+            self.train_loader.set_epoch(epoch)  # device-GT stream
         meter = AverageMeter()
         iters_per_epoch = len(self.train_loader)
         for it, batch in enumerate(self.train_loader):

@@ -45,6 +45,8 @@ def main():
                     help="synthetic dataset length per epoch")
     ap.add_argument("--num-workers", type=int, default=4)
     ap.add_argument("--no-sync-bn", action="store_true")
+    ap.add_argument("--device-data", type=int, default=0, metavar="STEPS",
+                    help="use the on-device GT generator for STEPS batches/epoch")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -76,7 +78,8 @@ def main():
     trainer = cls(opt, config, train_ds, val_ds, rank=rank,
                   local_rank=local_rank, world_size=world_size,
                   sync_bn=not args.no_sync_bn and world_size > 1,
-                  num_workers=args.num_workers, checkpoint_dir=args.ckpt_dir)
+                  num_workers=args.num_workers, checkpoint_dir=args.ckpt_dir,
+                  device_synth_steps=args.device_data)
     if args.resume:
         trainer.resume(args.resume)
     trainer.fit(args.epochs)

@@ -154,3 +154,22 @@ def test_device_gt_matches_oracle():
     got = create_heatmaps_device(joints, masks, config).cpu().numpy()
     assert got.shape == want.shape == (N, config.num_layers, h, w)
     np.testing.assert_allclose(got, want, atol=2e-5, rtol=1e-4)
+
+
+@pytest.mark.gpu
+def test_device_gt_loader_stream():
+    import torch
+    from improved_body_parts_amd.config import GetConfig
+    from improved_body_parts_amd.data import DeviceGTSyntheticLoader
+
+    config = GetConfig("Canonical")
+    loader = DeviceGTSyntheticLoader(config, batch_size=3, steps_per_epoch=2,
+                                     dtype=torch.bfloat16)
+    batches = list(loader)
+    assert len(batches) == 2
+    img, mm, hm = batches[0]
+    assert img.shape == (3, 512, 512, 3) and img.is_cuda
+    assert hm.shape == (3, config.num_layers, 128, 128)
+    assert mm.shape == (3, 1, 128, 128)
+    assert torch.isfinite(hm.float()).all()
+    assert float(hm.float().max()) <= 1.0 + 1e-3
