@@ -85,13 +85,6 @@ def conv_fwd(x, weight, stride, padding, dilation,
     Conv+BN+LeakyReLU module is ONE kernel on the inference path."""
     if not _supported(x, weight, stride, padding, dilation):
         return None
-    # plain large-M 1x1 with no fused epilogue is a bare GEMM — the library
-    # GEMM wins there (285 vs 206 TF measured, profiles/conv_shapes_vs_miopen);
-    # every fused case stays on the MFMA kernel (one kernel beats GEMM+2 ops)
-    if (weight.shape[2] == 1 and weight.shape[3] == 1 and scale is None
-            and residual is None and not act
-            and x.shape[0] * x.shape[2] * x.shape[3] >= (1 << 16)):
-        return None
     ext = hip_extension()
     if not hasattr(ext, "conv_mfma_fwd"):
         return None
@@ -112,9 +105,6 @@ def conv_fwd(x, weight, stride, padding, dilation,
 def conv_dgrad(dy, weight, x_shape, stride, padding, dilation):
     if not _supported(dy, weight, stride, padding, dilation, for_grad=True):
         return None
-    if (weight.shape[2] == 1 and weight.shape[3] == 1
-            and dy.shape[0] * dy.shape[2] * dy.shape[3] >= (1 << 16)):
-        return None  # bare 1x1 GEMM: library wins (see conv_fwd note)
     ext = hip_extension()
     if not hasattr(ext, "conv_mfma_fwd"):
         return None
