@@ -124,6 +124,36 @@ __global__ void se_reduce_kernel(const T* __restrict__ a, const T* __restrict__ 
 }
 
 // y = x * s[n][c] (+ optional add[n][c] broadcast)
+
+// fused SE gate: s = sigmoid(W2 @ leaky(W1 @ pooled + b1) + b2) — replaces
+// two library GEMM launches + bias/activation elementwise on the inference
+// path (the FCs are 256x16: far below any GEMM library's useful size).
+template <typename T>
+__global__ void se_gate_kernel(const float* __restrict__ pooled,
+                               const T* __restrict__ w1, const T* __restrict__ b1,
+                               const T* __restrict__ w2, const T* __restrict__ b2,
+                               float* __restrict__ s, int C, int CH,
+                               float slope) {
+  __shared__ float pl[768 + 64];
+  const int n = blockIdx.x;
+  const int tid = threadIdx.x;
+  for (int c = tid; c < C; c += blockDim.x) pl[c] = pooled[(long long)n * C + c];
+  __syncthreads();
+  if (tid < CH) {
+    float acc = ldf(b1 + tid);
+    const T* row = w1 + (long long)tid * C;
+    for (int c = 0; c < C; ++c) acc += ldf(row + c) * pl[c];
+    pl[C + tid] = leaky(acc, slope);
+  }
+  __syncthreads();
+  for (int c = tid; c < C; c += blockDim.x) {
+    float acc = ldf(b2 + c);
+    const T* row = w2 + (long long)c * CH;
+    for (int j = 0; j < CH; ++j) acc += ldf(row + j) * pl[C + j];
+    s[(long long)n * C + c] = 1.f / (1.f + __expf(-acc));
+  }
+}
+
 template <typename T>
 __global__ void se_scale_kernel(const T* __restrict__ x, const float* __restrict__ s,
                                 const float* __restrict__ addc, T* __restrict__ y,
@@ -226,6 +256,30 @@ Tensor se_reduce(const Tensor& a, const c10::optional<Tensor>& b, int64_t N,
     }
   });
   return out;
+}
+
+Tensor se_gate(const Tensor& pooled, const Tensor& w1, const Tensor& b1,
+               const Tensor& w2, const Tensor& b2, double slope) {
+  TORCH_CHECK(pooled.is_cuda() && pooled.dim() == 2 &&
+              pooled.scalar_type() == at::ScalarType::Float);
+  int64_t N = pooled.size(0), C = pooled.size(1);
+  int64_t CH = w1.size(0);
+  TORCH_CHECK(C <= 768 && CH <= 64, "se_gate LDS layout limit");
+  TORCH_CHECK(w1.is_contiguous() && w2.is_contiguous());
+  Tensor s = torch::empty({N, C}, pooled.options());
+  dim3 block(256), grid((unsigned)N);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+      w1.scalar_type(), "se_gate", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(ibp::se_gate_kernel<T>, grid, block, 0, cur_stream2(),
+                       pooled.data_ptr<float>(),
+                       reinterpret_cast<const T*>(w1.data_ptr()),
+                       reinterpret_cast<const T*>(b1.data_ptr()),
+                       reinterpret_cast<const T*>(w2.data_ptr()),
+                       reinterpret_cast<const T*>(b2.data_ptr()),
+                       s.data_ptr<float>(), (int)C, (int)CH, (float)slope);
+  });
+  return s;
 }
 
 Tensor se_scale(const Tensor& x, const Tensor& s, const c10::optional<Tensor>& addc,

@@ -110,6 +110,16 @@ class _SeScaleFn(torch.autograd.Function):
 
 
 def se_layer_hip(x, fc1, fc2):
+    if not torch.is_grad_enabled():
+        # inference: GAP + one fused gate kernel + scale (3 launches instead
+        # of ~7; the 256x16 FCs are far below useful GEMM-library sizes)
+        ext = hip_extension()
+        xc = x.contiguous(memory_format=_CL)
+        n, c, h_, w_ = xc.shape
+        pooled = ext.se_reduce(xc, None, n, h_ * w_, c) / float(h_ * w_)
+        s = ext.se_gate(pooled, fc1.weight, fc1.bias, fc2.weight, fc2.bias,
+                        LEAKY_SLOPE)
+        return ext.se_scale(xc, s, None, n, h_ * w_, c).permute(0, 3, 1, 2)
     pooled = _SeGapFn.apply(x)                       # [N, C] fp32
     h = F.leaky_relu(fc1(pooled.to(fc1.weight.dtype)), LEAKY_SLOPE)
     s = torch.sigmoid(fc2(h))

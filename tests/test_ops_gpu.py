@@ -382,3 +382,22 @@ def test_network_bf16_trains():
         losses.append(float(loss))
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]
+
+
+def test_se_gate_fused_inference():
+    """Fused GAP->FC->leaky->FC->sigmoid->scale vs the eager module chain."""
+    from improved_body_parts_amd.models import SELayer
+    torch.manual_seed(3)
+    for c, dtype in [(256, torch.bfloat16), (128, torch.float32),
+                     (768, torch.bfloat16)]:
+        se = SELayer(c).cuda().to(dtype)
+        x = torch.randn(3, c, 16, 16, device="cuda").to(dtype) \
+            .contiguous(memory_format=CL)
+        with torch.no_grad():
+            y = se(x)  # fused path (grad disabled)
+        pooled = x.float().mean(dim=(2, 3))
+        h = F.leaky_relu(pooled @ se.fc[0].weight.float().t()
+                         + se.fc[0].bias.float(), 0.01)
+        s = torch.sigmoid(h @ se.fc[2].weight.float().t() + se.fc[2].bias.float())
+        ref = x.float() * s.view(3, c, 1, 1)
+        _assert_rel(y, ref, 2e-2 if dtype == torch.bfloat16 else 1e-4, f"se C={c}")
