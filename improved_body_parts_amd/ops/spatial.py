@@ -119,7 +119,7 @@ def se_layer_hip(x, fc1, fc2):
         pooled = ext.se_reduce(xc, None, n, h_ * w_, c) / float(h_ * w_)
         s = ext.se_gate(pooled, fc1.weight, fc1.bias, fc2.weight, fc2.bias,
                         LEAKY_SLOPE)
-        return ext.se_scale(xc, s, None, n, h_ * w_, c).permute(0, 3, 1, 2)
+        return ext.se_scale(xc, s, None, n, h_ * w_, c)
     pooled = _SeGapFn.apply(x)                       # [N, C] fp32
     h = F.leaky_relu(fc1(pooled.to(fc1.weight.dtype)), LEAKY_SLOPE)
     s = torch.sigmoid(fc2(h))
