@@ -62,7 +62,9 @@ class GradReducer:
         if not params:
             raise ValueError("model has no trainable parameters")
         device = params[0].device
-        comm_dtype = comm_dtype or params[0].dtype
+        # comm_dtype is accepted for API stability but gradients must live in
+        # each parameter's own dtype (the .grad views alias the flat buckets)
+        del comm_dtype
 
         if self.world_size > 1 and broadcast_parameters:
             for p in params:
@@ -71,18 +73,23 @@ class GradReducer:
                 if b.dtype.is_floating_point or b.dtype in (torch.int64, torch.int32):
                     dist.broadcast(b.data, src=0, group=self.group)
 
-        # reverse order ≈ order gradients become ready during backward
+        # reverse order ≈ order gradients become ready during backward.
+        # Buckets are PER-DTYPE: a bf16 model still carries fp32 BN affine
+        # parameters, and a parameter's .grad view must match its dtype.
         cap = int(bucket_cap_mb * 1024 * 1024)
         self.buckets: List[Bucket] = []
-        cur, cur_bytes = [], 0
+        cur: dict = {}
+        cur_bytes: dict = {}
         for p in reversed(params):
-            cur.append(p)
-            cur_bytes += p.numel() * p.element_size()
-            if cur_bytes >= cap:
-                self.buckets.append(Bucket(cur, device, comm_dtype))
-                cur, cur_bytes = [], 0
-        if cur:
-            self.buckets.append(Bucket(cur, device, comm_dtype))
+            dt = p.dtype
+            cur.setdefault(dt, []).append(p)
+            cur_bytes[dt] = cur_bytes.get(dt, 0) + p.numel() * p.element_size()
+            if cur_bytes[dt] >= cap:
+                self.buckets.append(Bucket(cur[dt], device, dt))
+                cur[dt], cur_bytes[dt] = [], 0
+        for dt, ps in cur.items():
+            if ps:
+                self.buckets.append(Bucket(ps, device, dt))
 
         self._param_bucket = {}
         for b in self.buckets:

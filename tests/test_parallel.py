@@ -160,3 +160,33 @@ def test_reduce_tensor_single_process():
     from improved_body_parts_amd.parallel import reduce_tensor
     t = torch.tensor(3.0)
     assert float(reduce_tensor(t)) == 3.0
+
+
+def _mixed_dtype_worker(rank, world, tmpdir):
+    import torch.distributed as dist
+    from improved_body_parts_amd.parallel import GradReducer
+    dist.init_process_group("gloo", init_method=f"file://{tmpdir}/mixed_rdv",
+                            rank=rank, world_size=world)
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Linear(8, 4))
+    model[0].double()  # mixed parameter dtypes (the bf16-model + fp32-BN case)
+    reducer = GradReducer(model, bucket_cap_mb=1e-4)  # force many buckets
+    for p in model.parameters():
+        assert p.grad is not None and p.grad.dtype == p.dtype
+    reducer.zero_grad()
+    x = torch.randn(4, 8)
+    y = model[1](model[0](x.double()).float()).sum() * (rank + 1)
+    y.backward()
+    reducer.finalize()
+    g = model[0].weight.grad.clone()
+    dist.all_reduce(g.div_(1))  # compare against the mean both ranks hold
+    assert torch.allclose(model[0].weight.grad * world, g, atol=1e-9)
+    dist.destroy_process_group()
+
+
+def test_grad_reducer_mixed_dtypes(tmp_path):
+    """Per-dtype buckets: fp64+fp32 params reduce correctly over gloo (the
+    CPU stand-in for the bf16-weights + fp32-BN layout on MI355X)."""
+    import torch.multiprocessing as mp
+    mp.spawn(_mixed_dtype_worker, args=(2, str(tmp_path)), nprocs=2,
+             join=True)
