@@ -111,8 +111,7 @@ def main():
             hm = hm[None].expand(batch, -1, -1, -1).contiguous()
             batches.append(tuple(t.to(device=device, dtype=dtype) for t in (img, mm, hm)))
 
-        def step(i):
-            b = batches[i % len(batches)]
+        def train_step(b):
             if reducer is not None:
                 reducer.zero_grad()
             else:
@@ -125,6 +124,27 @@ def main():
                 reducer.finalize()
             optimizer.step()
             return loss
+
+        if args.graph and use_cuda and not distributed:
+            # whole-step hipGraph: fwd + focal-L2 + bwd + fused SGD captured
+            # once, replayed per step (there is no host sync inside the step)
+            static = batches[0]
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    train_step(static)
+            torch.cuda.current_stream().wait_stream(s)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_loss = train_step(static)
+
+            def step(i):
+                graph.replay()
+                return static_loss
+        else:
+            def step(i):
+                return train_step(batches[i % len(batches)])
 
         metric_name = "train_images_per_sec"
         vs_baseline = None
