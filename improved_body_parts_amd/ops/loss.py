@@ -5,6 +5,20 @@ import torch
 
 from ._backend import hip_extension
 
+_NW_CACHE = {}
+
+
+def _nstack_weight_tensor(nstack_weight, device):
+    """Constant per config — cached on device (a per-call as_tensor would do a
+    pageable H2D copy, which hipGraph capture forbids)."""
+    key = (tuple(nstack_weight), device)
+    t = _NW_CACHE.get(key)
+    if t is None:
+        t = torch.as_tensor(list(nstack_weight), dtype=torch.float32,
+                            device=device)
+        _NW_CACHE[key] = t
+    return t
+
 
 class FocalL2Fn(torch.autograd.Function):
     @staticmethod
@@ -16,8 +30,7 @@ class FocalL2Fn(torch.autograd.Function):
         mask = mask.contiguous()
         sums = ext.focal_l2_fwd(pred, gt, mask, heat_start, bkg_start, gamma,
                                 mtw, ktw, alpha, beta)
-        nw = torch.as_tensor(nstack_weight, dtype=torch.float32,
-                             device=pred.device)
+        nw = _nstack_weight_tensor(nstack_weight, pred.device)
         loss = (sums * nw).sum() / nw.sum()
         ctx.save_for_backward(pred, gt, mask, nw)
         ctx.conf = (heat_start, bkg_start, gamma, mtw, ktw, alpha, beta)
