@@ -21,6 +21,10 @@ def _build_table(batch, device):
             rows.append((pp, gp, mp, wp, off, cnt))
             off += cnt
     t = torch.tensor(rows, dtype=torch.int64)
+    if device.type == "cuda":
+        # pinned staging: a pageable async H2D is a no-op under hipGraph
+        # capture (the kernel then reads a dangling table pointer)
+        t = t.pin_memory()
     return t.to(device, non_blocking=True)
 
 
@@ -30,10 +34,13 @@ def fused_sgd_step(batch, lr, momentum, weight_decay):
         return
     ext = hip_extension()
     device = batch[0][0].device
-    key = tuple(id(p) for p, _, _, _ in batch) + \
-        tuple(b[1].data_ptr() for b in batch[:1])
+    key = tuple(id(p) for p, _, _, _ in batch)
     entry = _table_cache.get(key)
-    ptrs = tuple(b[0].data_ptr() for b in batch) + tuple(b[1].data_ptr() for b in batch)
+    # freshness covers EVERY pointer the table embeds — params, grads,
+    # momentum and master buffers (set_to_none grads move between steps;
+    # a stale row is a dangling device pointer)
+    ptrs = tuple(t.data_ptr() if t is not None else 0
+                 for row in batch for t in row)
     if entry is None or entry[0] != ptrs:
         table = _build_table(batch, device)
         _table_cache[key] = (ptrs, table)
