@@ -116,8 +116,11 @@ def main():
                 reducer.zero_grad()
             else:
                 # None grads skip ~800 fill launches AND turn the first
-                # autograd accumulation per tensor into an assignment
-                optimizer.zero_grad(set_to_none=True)
+                # autograd accumulation per tensor into an assignment.
+                # Graph mode needs STATIC grad buffers instead: with stable
+                # addresses the fused-SGD pointer table never rebuilds, so
+                # nothing capture-hostile runs inside the capture.
+                optimizer.zero_grad(set_to_none=not args.graph)
             loss = model(b)
             loss.backward()
             if reducer is not None:
