@@ -13,7 +13,6 @@ Storage format: same two-group layout the reference writes
 from __future__ import annotations
 
 import json
-import random
 
 import numpy as np
 import torch

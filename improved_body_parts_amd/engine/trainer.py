@@ -26,7 +26,6 @@ import os
 import time
 
 import torch
-import torch.distributed as dist
 from torch.utils.data import DataLoader
 from torch.utils.data.distributed import DistributedSampler
 

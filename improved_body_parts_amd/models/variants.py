@@ -29,8 +29,7 @@ import torch
 from torch import nn
 
 from .. import ops
-from .layers import (LEAKY_SLOPE, BasicResidual, Conv, DilatedConv, Hourglass,
-                     Residual, SELayer)
+from .layers import LEAKY_SLOPE, Conv, Hourglass, Residual, SELayer
 from .posenet import Merge, PoseNet
 
 

@@ -9,8 +9,6 @@ from __future__ import annotations
 
 import glob
 import os
-import shlex
-import subprocess
 import sys
 
 PKG_DIR = os.path.dirname(os.path.abspath(__file__))

@@ -1,7 +1,6 @@
 """Device post-processing wrappers (peaks, limb scoring)."""
 from __future__ import annotations
 
-import torch
 
 from ._backend import hip_extension
 

@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import torch
 import torch.distributed as dist
-import torch.nn.functional as F
 from torch import nn
 
 
