@@ -40,7 +40,10 @@ class TrainingOpt:
         self.warmup_epochs = 3
         self.lr_decay_every = 15       # epochs between x0.2 decays (reference train_distributed.py:382-400)
         self.lr_decay_factor = 0.2
-        self.loss_explosion_thre = 2e5  # drop batches above this (reference train_distributed.py:259-261)
+        # drop batches above this (reference train_distributed.py:259-261 used
+        # 2e5 for sparse COCO crops; dense synthetic scenes start near 2e5 at
+        # random init, so the guard sits well above the healthy range)
+        self.loss_explosion_thre = 1e7
         for k, v in overrides.items():
             if not hasattr(self, k):
                 raise AttributeError(f"unknown TrainingOpt field {k!r}")
