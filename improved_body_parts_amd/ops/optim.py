@@ -48,3 +48,8 @@ def fused_sgd_step(batch, lr, momentum, weight_decay):
         table = entry[1]
     dtype_tag = 0 if batch[0][0].dtype == torch.bfloat16 else 1
     ext.fused_sgd(table, lr, momentum, weight_decay, dtype_tag)
+    # the kernel writes parameters in place without going through ATen, so
+    # autograd version counters never tick — bump them or version-keyed
+    # caches (the conv weight packs) serve stale data forever
+    for p, _, _, _ in batch:
+        torch.autograd.graph.increment_version(p)
