@@ -22,12 +22,14 @@ struct SgdChunk {
   float* master;      // fp32 master (nullptr when param is fp32)
   long long offset;   // element offset of this chunk within the tensor
   int count;          // elements in this chunk
+  int dtype;          // 0 = bf16, 1 = fp32 — PER CHUNK: a bf16 model still
+                      // carries fp32 BN affine parameters, and one global tag
+                      // would reinterpret (and corrupt) their memory
 };
 
 template <typename T>
-__global__ void fused_sgd_kernel(const SgdChunk* __restrict__ chunks,
-                                 float lr, float mu, float wd) {
-  const SgdChunk ck = chunks[blockIdx.x];
+__device__ __forceinline__ void sgd_chunk_update(const SgdChunk& ck, float lr,
+                                                 float mu, float wd) {
   T* p = reinterpret_cast<T*>(ck.param) + ck.offset;
   const T* g = reinterpret_cast<const T*>(ck.grad) + ck.offset;
   float* m = ck.momentum + ck.offset;
@@ -43,14 +45,25 @@ __global__ void fused_sgd_kernel(const SgdChunk* __restrict__ chunks,
   }
 }
 
+__global__ void fused_sgd_kernel(const SgdChunk* __restrict__ chunks,
+                                 float lr, float mu, float wd) {
+  const SgdChunk ck = chunks[blockIdx.x];
+  if (ck.dtype == 0) {
+    sgd_chunk_update<__hip_bfloat16>(ck, lr, mu, wd);
+  } else {
+    sgd_chunk_update<float>(ck, lr, mu, wd);
+  }
+}
+
 }  // namespace ibp
 
 using torch::Tensor;
 
 // chunk table packed on host into an int64 tensor [n][6]:
-// (param_ptr, grad_ptr, momentum_ptr, master_ptr, offset, count)
+// (param_ptr, grad_ptr, momentum_ptr, master_ptr, offset,
+//  count | dtype << 32)    with dtype 0 = bf16, 1 = fp32
 void fused_sgd(const Tensor& chunk_table, double lr, double momentum,
-               double weight_decay, int64_t dtype_tag) {
+               double weight_decay, int64_t dtype_tag /* unused */) {
   TORCH_CHECK(chunk_table.is_cuda() && chunk_table.dtype() == torch::kInt64);
   int n = (int)chunk_table.size(0);
   static_assert(sizeof(ibp::SgdChunk) == 6 * 8, "chunk layout mismatch");
@@ -58,12 +71,6 @@ void fused_sgd(const Tensor& chunk_table, double lr, double momentum,
       reinterpret_cast<const ibp::SgdChunk*>(chunk_table.data_ptr<int64_t>());
   auto stream = at::hip::getCurrentHIPStream().stream();
   dim3 block(256), grid(n);
-  if (dtype_tag == 0) {  // bf16 params
-    hipLaunchKernelGGL(ibp::fused_sgd_kernel<__hip_bfloat16>, grid, block, 0,
-                       stream, chunks, (float)lr, (float)momentum,
-                       (float)weight_decay);
-  } else {  // fp32 params
-    hipLaunchKernelGGL(ibp::fused_sgd_kernel<float>, grid, block, 0, stream,
-                       chunks, (float)lr, (float)momentum, (float)weight_decay);
-  }
+  hipLaunchKernelGGL(ibp::fused_sgd_kernel, grid, block, 0, stream, chunks,
+                     (float)lr, (float)momentum, (float)weight_decay);
 }

@@ -16,9 +16,12 @@ def _build_table(batch, device):
         off = 0
         pp, gp, mp = p.data_ptr(), g.data_ptr(), m.data_ptr()
         wp = master.data_ptr() if master is not None else 0
+        # per-chunk dtype tag in the high half of the count word (a bf16
+        # model still has fp32 BN affine parameters)
+        dtag = 0 if p.dtype == torch.bfloat16 else 1
         while off < numel:
             cnt = min(CHUNK, numel - off)
-            rows.append((pp, gp, mp, wp, off, cnt))
+            rows.append((pp, gp, mp, wp, off, cnt | (dtag << 32)))
             off += cnt
     t = torch.tensor(rows, dtype=torch.int64)
     if device.type == "cuda":

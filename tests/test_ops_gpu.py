@@ -401,3 +401,29 @@ def test_se_gate_fused_inference():
         s = torch.sigmoid(h @ se.fc[2].weight.float().t() + se.fc[2].bias.float())
         ref = x.float() * s.view(3, c, 1, 1)
         _assert_rel(y, ref, 2e-2 if dtype == torch.bfloat16 else 1e-4, f"se C={c}")
+
+
+def test_fused_sgd_mixed_dtypes_vs_torch():
+    """HIP fused SGD on a MIXED bf16+fp32 parameter set vs torch.optim.SGD
+    (one global dtype tag once corrupted the fp32 BN affine params)."""
+    from improved_body_parts_amd.engine import FusedSGD
+    torch.manual_seed(0)
+    w_bf = torch.nn.Parameter(torch.randn(70000, device="cuda").bfloat16())
+    w_fp = torch.nn.Parameter(torch.randn(333, device="cuda"))
+    ref_bf = torch.nn.Parameter(w_bf.detach().float().clone())
+    ref_fp = torch.nn.Parameter(w_fp.detach().clone())
+    opt = FusedSGD([w_bf, w_fp], lr=0.1, momentum=0.9, weight_decay=0.01)
+    ref = torch.optim.SGD([ref_bf, ref_fp], lr=0.1, momentum=0.9,
+                          weight_decay=0.01)
+    for step in range(4):
+        g = torch.randn(70000, device="cuda")
+        g2 = torch.randn(333, device="cuda")
+        w_bf.grad = g.bfloat16()
+        w_fp.grad = g2.clone()
+        ref_bf.grad = g.bfloat16().float()  # same quantized grad
+        ref_fp.grad = g2.clone()
+        opt.step()
+        ref.step()
+    _assert_rel(w_fp, ref_fp, 1e-6, "fp32 params")
+    _assert_rel(w_bf.float(), ref_bf, 1e-2, "bf16 params (fp32 master)")
+    assert torch.isfinite(w_fp).all() and torch.isfinite(w_bf.float()).all()
