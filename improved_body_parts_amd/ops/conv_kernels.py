@@ -21,11 +21,11 @@ from ._backend import hip_extension
 _CL = torch.channels_last
 
 # shapes supported by the MFMA path: stride-1 any dilation, and stride-2 fwd
-_DISABLE = os.environ.get("IBP_AMD_DISABLE_MFMA_CONV") == "1"
+DISABLE = os.environ.get("IBP_AMD_DISABLE_MFMA_CONV") == "1"
 
 
 def _supported(x, weight, stride, padding, dilation, for_grad=False):
-    if _DISABLE:
+    if DISABLE:
         return False
     if x.dtype != torch.bfloat16:
         return False
