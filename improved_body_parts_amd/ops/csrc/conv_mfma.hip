@@ -329,10 +329,17 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
   const int BN = (Cout > 64 && padded128 <= padded64) ? 128 : 64;
   const int ntiles = p.n_mtiles * (int)((Cout + BN - 1) / BN);
   const int nk_total = (p.K + ibp::BK - 1) / ibp::BK;
-  // split K on small grids so the 256-CU chip stays filled (~2 blocks/CU)
+  // split K on small grids so the 256-CU chip stays filled (~2 blocks/CU).
+  // Normalise so EVERY slice has work: with raw splitk=7 over nk_total=8 the
+  // last 3 slices get no chunks, return early, and the combine kernel then
+  // sums their UNINITIALISED workspace slices (silent garbage whenever the
+  // allocator hands back dirty memory — found as exploding gradients through
+  // the scale-2 Merge convs, scripts/diag_splitk.py).
   int splitk = 1;
   if (ntiles < 384 && nk_total > 1) {
     splitk = std::min((int)nk_total, (384 + ntiles - 1) / ntiles);
+    int nk_chunk = ((int)nk_total + splitk - 1) / splitk;
+    splitk = ((int)nk_total + nk_chunk - 1) / nk_chunk;
   }
   p.splitk = splitk;
   Tensor ws;

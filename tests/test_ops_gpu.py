@@ -427,3 +427,23 @@ def test_fused_sgd_mixed_dtypes_vs_torch():
     _assert_rel(w_fp, ref_fp, 1e-6, "fp32 params")
     _assert_rel(w_bf.float(), ref_bf, 1e-2, "bf16 params (fp32 master)")
     assert torch.isfinite(w_fp).all() and torch.isfinite(w_bf.float()).all()
+
+
+def test_mfma_conv_ragged_splitk():
+    """Shapes where splitk does not divide the K-chunk count: trailing slices
+    must still be accounted (uninitialised-workspace regression,
+    scripts/diag_splitk.py). Run twice so the allocator serves DIRTY memory."""
+    from improved_body_parts_amd.ops import conv_kernels
+    for _ in range(2):
+        for (n, cin, cout, hw) in [(8, 512, 50, 32), (8, 512, 64, 32),
+                                   (8, 512, 128, 32), (8, 1024, 50, 32)]:
+            torch.manual_seed(1)
+            x = torch.randn(n, cin, hw, hw, device="cuda").bfloat16() \
+                .contiguous(memory_format=CL)
+            # dirty the allocator pool so empty() workspaces are non-zero
+            junk = torch.full((64 << 20,), 3.3e7, device="cuda")
+            del junk
+            w = (torch.randn(cout, cin, 1, 1, device="cuda") * 0.05).bfloat16()
+            y = conv_kernels.conv_fwd(x, w, (1, 1), (0, 0), (1, 1))
+            ref = F.conv2d(x.float(), w.float())
+            _assert_rel(y, ref, 1e-2, f"ragged splitk {(n, cin, cout, hw)}")
