@@ -218,12 +218,14 @@ def find_peaks(heatmap_avg, params, config: CanonicalConfig, max_peaks=512):
 # --------------------------------------------------------------------------
 
 def _limb_scores_host(paf_np, pa, pb, mid_num, thre2):
-    """CPU scoring of one candidate segment (same math as limb_score_kernel)."""
+    """CPU scoring of one candidate segment (same math as limb_score_kernel).
+    Short limbs sample fewer points (reference evaluate.py:228)."""
     H, W = paf_np.shape
     ax, ay, bx, by = pa[0], pa[1], pb[0], pb[1]
     norm = math.hypot(bx - ax, by - ay) + 1e-9
-    xs = np.round(np.linspace(ax, bx, mid_num)).astype(int).clip(0, W - 1)
-    ys = np.round(np.linspace(ay, by, mid_num)).astype(int).clip(0, H - 1)
+    mn = min(int(round(norm)) + 1, mid_num)
+    xs = np.round(np.linspace(ax, bx, mn)).astype(int).clip(0, W - 1)
+    ys = np.round(np.linspace(ay, by, mn)).astype(int).clip(0, H - 1)
     v = paf_np[ys, xs]
     mean = float(v.mean()) + min(0.5 * H / norm - 1.0, 0.0)
     return mean, float((v > thre2).mean()), float(norm)

@@ -100,8 +100,11 @@ __global__ void limb_score_kernel(
     const T* plane = paf + (long long)limb * H * W;
     float sum = 0.f;
     int pass = 0;
-    for (int k = 0; k < mid_num; ++k) {
-      float t = mid_num == 1 ? 0.f : (float)k / (mid_num - 1);
+    // short limbs sample fewer points (reference evaluate.py:228:
+    // mid_num = min(round(norm)+1, param mid_num))
+    const int mn = min((int)roundf(len) + 1, mid_num);
+    for (int k = 0; k < mn; ++k) {
+      float t = mn == 1 ? 0.f : (float)k / (mn - 1);
       int x = (int)roundf(ax + t * dx);
       int y = (int)roundf(ay + t * dy);
       x = min(max(x, 0), W - 1);
@@ -110,12 +113,12 @@ __global__ void limb_score_kernel(
       sum += v;
       if (v > thre2) ++pass;
     }
-    float mean = sum / mid_num;
+    float mean = sum / mn;
     // distance prior of the reference (evaluate.py:240): penalise limbs longer
     // than half the image height
     float prior = fminf(0.5f * H / len - 1.f, 0.f);
     scores[i * 3 + 0] = mean + prior;
-    scores[i * 3 + 1] = (float)pass / mid_num;
+    scores[i * 3 + 1] = (float)pass / mn;
     scores[i * 3 + 2] = len;
   }
 }

@@ -247,3 +247,53 @@ def test_grouping_matches_reference_evaluate():
     np.testing.assert_allclose(o[:, :18, 0], r[:, :18, 0])   # part assignment
     np.testing.assert_allclose(o[:, -1, 0], r[:, -1, 0])     # part counts
     np.testing.assert_allclose(o[:, -2, 0], r[:, -2, 0], rtol=1e-5)  # scores
+
+
+@pytest.mark.parametrize("remove_recon", [0, 1])
+def test_grouping_matches_reference_crowded(remove_recon):
+    """Crowded random scenes (overlapping people, short limbs, missing parts)
+    through both greedy-assembly implementations, including the competition-
+    resolution branch (remove_recon=1)."""
+    ref_eval = _import_reference_evaluate()
+    from improved_body_parts_amd.config import GetConfig, InferenceParams
+    from improved_body_parts_amd.data import sample_people
+    from improved_body_parts_amd.engine.inference import (find_connections,
+                                                          find_peaks,
+                                                          find_people)
+    config = GetConfig("Canonical")
+    params, _ = InferenceParams().as_params_dict()
+    params["remove_recon"] = remove_recon
+    rng = np.random.default_rng(123 + remove_recon)
+    H = W = 320
+    ppl = sample_people(rng, W, H, max_people=5)
+    people = []
+    for p in ppl:
+        d = {j: (float(p[j, 0]), float(p[j, 1])) for j in range(18)
+             if p[j, 2] < 2 and 0 <= p[j, 0] < W and 0 <= p[j, 1] < H}
+        if d:
+            people.append(d)
+    heat, paf = _synth_scene(config, people, H=H, W=W)
+    all_peaks = find_peaks(torch.from_numpy(heat), params, config)
+    o_conn, o_sp = find_connections(all_peaks, torch.from_numpy(paf), H,
+                                    params, config)
+    r_conn, r_sp = ref_eval.find_connections(all_peaks, paf, H, dict(params))
+    assert o_sp == r_sp
+    for k in range(len(r_conn)):
+        r = np.asarray(r_conn[k], dtype=np.float64).reshape(-1, 6)
+        o = np.asarray(o_conn[k], dtype=np.float64).reshape(-1, 6)
+        assert o.shape == r.shape, f"limb {k}: {o.shape} vs {r.shape}"
+        if len(r):
+            o = o[np.lexsort((o[:, 1], o[:, 0]))]
+            r = r[np.lexsort((r[:, 1], r[:, 0]))]
+            np.testing.assert_array_equal(o[:, :2], r[:, :2])
+            np.testing.assert_allclose(o[:, 2], r[:, 2], rtol=1e-5)
+    o_sub, o_cand = find_people(o_conn, o_sp, all_peaks, params, config)
+    r_sub, r_cand = ref_eval.find_people(r_conn, r_sp, all_peaks, dict(params))
+    assert len(o_sub) == len(r_sub)
+    if len(r_sub):
+        o = np.asarray(o_sub)
+        r = np.asarray(r_sub)
+        o = o[np.argsort(o[:, :18, 0].max(axis=1))]
+        r = r[np.argsort(r[:, :18, 0].max(axis=1))]
+        np.testing.assert_allclose(o[:, :18, 0], r[:, :18, 0])
+        np.testing.assert_allclose(o[:, -2, 0], r[:, -2, 0], rtol=1e-5)
