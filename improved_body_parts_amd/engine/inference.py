@@ -122,7 +122,6 @@ def predict(image, model, config: CanonicalConfig, params=None, model_params=Non
 
     multiplier = [s * model_params["boxsize"] / H for s in params["scale_search"]]
     rotations = params.get("rotation_search", [0.0])
-    stride = model_params["stride"]
     pad_to = model_params["max_downsample"]
     pad_value = model_params["padValue"] / 255.0
 

@@ -70,8 +70,10 @@ def main():
         val_ds = SyntheticPoseDataset(config, length=max(args.batch * 4, 16),
                                       seed=101)
     else:
+        from improved_body_parts_amd.config import COCOSourceConfig
         from improved_body_parts_amd.data import MyDataset
-        train_ds = MyDataset(config, args.data, shuffle=True, augment=True)
+        train_ds = MyDataset(config, COCOSourceConfig(args.data),
+                             shuffle=True, augment=True)
         val_ds = None
 
     cls = SWATrainer if args.swa else Trainer

@@ -206,3 +206,20 @@ def test_pipeline_device_vs_cpu_parity(config, params):
         for pc, pd in zip(sorted(sc), sorted(sd)):
             assert abs(pc[0] - pd[0]) < 0.05 and abs(pc[1] - pd[1]) < 0.05
             assert abs(pc[2] - pd[2]) < 0.01
+
+
+def test_predict_multiscale_and_rotation(config):
+    """Ensemble over 2 scales x 2 rotations runs and keeps shapes/finite."""
+    opt = TrainingOpt(nstack=1, batch_size=1)
+    model = NetworkEval(opt, config, bn=True).eval()
+    img = np.random.RandomState(2).rand(64, 64, 3).astype(np.float32)
+    p, mp = InferenceParams().as_params_dict()
+    p = dict(p)
+    mp = dict(mp)
+    mp["boxsize"] = 64
+    p["scale_search"] = [0.8, 1.0]
+    p["rotation_search"] = [0.0, 30.0]
+    with torch.no_grad():
+        heat, paf = predict(img, model, config, p, mp)
+    assert heat.shape[:2] == (64, 64)
+    assert torch.isfinite(heat).all() and torch.isfinite(paf).all()
