@@ -55,6 +55,14 @@ def _conv_wgrad(x, dy, w_shape, stride, padding, dilation):
     return dw
 
 
+def _tick_running_stats(bn_mod):
+    """The finalize kernel updates running stats in place without going
+    through ATen — tick their version counters so the eval-path folded-BN
+    cache (keyed on them) does not serve stale scale/shift."""
+    torch.autograd.graph.increment_version(bn_mod.running_mean)
+    torch.autograd.graph.increment_version(bn_mod.running_var)
+
+
 class ConvBnActFn(torch.autograd.Function):
     """y = leaky( bn( conv(x, w) ) (+ residual) ), all fused on device."""
 
@@ -115,6 +123,7 @@ class ConvBnActFn(torch.autograd.Function):
                         flat[:C], flat[C:], M, gamma.float(), beta.float(),
                         bn_mod.running_mean, bn_mod.running_var,
                         bn_mod.num_batches_tracked, mom, bn_mod.eps)
+                    _tick_running_stats(bn_mod)
                 else:
                     # single fused kernel chain: stats + colsum + per-channel
                     # epilogue + running-stat update (the Python mean/var/rsqrt
@@ -123,6 +132,7 @@ class ConvBnActFn(torch.autograd.Function):
                         y_conv, C, gamma.float(), beta.float(),
                         bn_mod.running_mean, bn_mod.running_var,
                         bn_mod.num_batches_tracked, mom, bn_mod.eps)
+                    _tick_running_stats(bn_mod)
             else:
                 mean = bn_mod.running_mean.float()
                 var = bn_mod.running_var.float()
