@@ -97,20 +97,42 @@ class MultiTaskLoss(nn.Module):
 class MultiTaskLossParallel(MultiTaskLoss):
     """Single-process / DataParallel-path loss (reference loss_model_parallel.py).
 
-    The reference's parallel variant defaults to the plain L2 loss and uses
-    gamma=2 for its focal option; batch division is done by the driver. We keep
-    the same semantics: this class defaults to gamma=2 focal and does NOT divide
-    by the batch size.
+    Semantics differ from the DDP loss and are preserved exactly (verified
+    against the reference in tests/test_reference_parity.py):
+      * default is PLAIN L2 (reference :68), focal (gamma=2, no alpha/beta,
+        no task weights) available via ``use_focal=True`` (:77-99);
+      * mask_miss is bilinear-interpolated WITHOUT the 0.5 threshold and
+        broadcast over channels — no per-channel task weighting;
+      * no batch division (the driver divides, reference train_parallel.py:145).
     """
 
-    def __init__(self, opt, config, **kwargs):
-        kwargs.setdefault("gamma", 2)
+    def __init__(self, opt, config, use_focal=False, gamma=2, **kwargs):
+        kwargs.setdefault("gamma", gamma)
         super().__init__(opt, config, **kwargs)
+        self.use_focal = use_focal
 
     def forward(self, pred_tuple, target_tuple):
         nstack = len(pred_tuple)
         loss_scales = []
         for i in range(5):
             pred = torch.stack([pred_tuple[j][i] for j in range(nstack)], dim=0)
-            loss_scales.append(self._loss_per_scale(pred, target_tuple) * self.scale_weight[i])
+            loss_scales.append(self._loss_per_scale(pred, target_tuple)
+                               * self.scale_weight[i])
         return sum(loss_scales) / sum(self.scale_weight)
+
+    def _loss_per_scale(self, pred, target):
+        size = pred.shape[-2:]
+        mask = F.interpolate(target[0].float(), size=size, mode="bilinear",
+                             align_corners=False)
+        gt = F.adaptive_avg_pool2d(target[1].float(), output_size=size)
+        pred = pred.float()
+        if self.use_focal:
+            st = torch.where(gt[None] >= 0.01, pred, 1 - pred)
+            factor = (1.0 - st) ** self.gamma
+            out = (pred - gt[None]) ** 2 * factor * mask[None]
+        else:
+            out = (pred - gt[None]) ** 2 * mask[None]
+        loss_nstack = out.sum(dim=(1, 2, 3, 4))
+        w = [loss_nstack[i] * self.nstack_weight[i]
+             for i in range(len(self.nstack_weight))]
+        return sum(w) / sum(self.nstack_weight)

@@ -75,19 +75,36 @@ def test_mask_miss_zero_region_excluded(small_config, small_opt):
     assert float(loss) == pytest.approx(0.0, abs=1e-6)
 
 
-def test_parallel_loss_no_batch_divide(small_config, small_opt):
-    """MultiTaskLossParallel keeps the reference semantics: same focal loss with
-    gamma=2 and no division by batch size (driver divides)."""
-    crit_p = MultiTaskLossParallel(small_opt, small_config)
-    crit_d = MultiTaskLoss(small_opt, small_config, gamma=2)
+def test_parallel_loss_reference_semantics(small_config, small_opt):
+    """MultiTaskLossParallel follows the reference's DataParallel-path
+    semantics exactly (loss_model_parallel.py): plain L2 by default with the
+    channel-broadcast unthresholded mask and NO batch division; focal option
+    uses gamma=2 without alpha/beta/task weights."""
+    import torch.nn.functional as F
+    crit = MultiTaskLossParallel(small_opt, small_config)
     nstack = small_opt.nstack
     torch.manual_seed(1)
     preds = [[torch.rand(2, 50, 32 // 2 ** s, 32 // 2 ** s) for s in range(5)]
              for _ in range(nstack)]
-    targ = (torch.ones(2, 1, 32, 32), torch.rand(2, 50, 32, 32))
-    lp = crit_p(preds, targ)
-    ld = crit_d(preds, targ)
-    assert torch.allclose(lp, ld * 2, rtol=1e-5)  # batch=2
+    mask = (torch.rand(2, 1, 32, 32) > 0.3).float()
+    gt = torch.rand(2, 50, 32, 32)
+    lp = crit(preds, (mask, gt))
+    # manual oracle
+    acc = 0.0
+    for i in range(5):
+        pred = torch.stack([preds[j][i] for j in range(nstack)], 0)
+        size = pred.shape[-2:]
+        m = F.interpolate(mask, size=size, mode="bilinear", align_corners=False)
+        g = F.adaptive_avg_pool2d(gt, output_size=size)
+        out = (pred - g[None]) ** 2 * m[None]
+        per_stack = out.sum(dim=(1, 2, 3, 4))
+        acc = acc + per_stack.mean() * small_opt.scale_weight[i]
+    want = acc / sum(small_opt.scale_weight)
+    assert torch.allclose(lp, want, rtol=1e-5)
+    # focal option differs from plain L2
+    lf = MultiTaskLossParallel(small_opt, small_config, use_focal=True)(
+        preds, (mask, gt))
+    assert not torch.allclose(lf, lp, rtol=1e-3)
 
 
 def test_plain_l2_and_l1_losses():

@@ -249,8 +249,9 @@ def test_grouping_matches_reference_evaluate():
     np.testing.assert_allclose(o[:, -2, 0], r[:, -2, 0], rtol=1e-5)  # scores
 
 
+@pytest.mark.parametrize("seed", [123, 77, 2024])
 @pytest.mark.parametrize("remove_recon", [0, 1])
-def test_grouping_matches_reference_crowded(remove_recon):
+def test_grouping_matches_reference_crowded(remove_recon, seed):
     """Crowded random scenes (overlapping people, short limbs, missing parts)
     through both greedy-assembly implementations, including the competition-
     resolution branch (remove_recon=1)."""
@@ -263,7 +264,7 @@ def test_grouping_matches_reference_crowded(remove_recon):
     config = GetConfig("Canonical")
     params, _ = InferenceParams().as_params_dict()
     params["remove_recon"] = remove_recon
-    rng = np.random.default_rng(123 + remove_recon)
+    rng = np.random.default_rng(seed + remove_recon)
     H = W = 320
     ppl = sample_people(rng, W, H, max_people=5)
     people = []
@@ -327,3 +328,22 @@ def test_posenet_checkpoint_and_forward_match_reference():
         assert a.shape == b.shape
         rel = float((a - b).norm() / (b.norm() + 1e-12))
         assert rel < 1e-5, f"scale {s}: rel {rel}"
+
+
+def test_parallel_loss_matches_reference(setups):
+    """MultiTaskLossParallel (plain-L2 default, no batch division) vs the
+    reference models/loss_model_parallel.py."""
+    _stub_ref_modules()
+    from models.loss_model_parallel import MultiTaskLossParallel as RefLoss
+    from improved_body_parts_amd.models import MultiTaskLossParallel
+    config, opt = setups
+    torch.manual_seed(3)
+    n, c = 2, config.num_layers
+    pred = [[torch.randn(n, c, 128 // (2 ** s), 128 // (2 ** s)) * 0.1
+             for s in range(5)] for _ in range(opt.nstack)]
+    mask = (torch.rand(n, 1, 128, 128) > 0.2).float()
+    gt = torch.rand(n, c, 128, 128)
+    ours = MultiTaskLossParallel(opt, config)(pred, (mask, gt))
+    ref = RefLoss(opt, config)(pred, (mask, gt))
+    assert torch.allclose(ours.float(), ref.float(), rtol=1e-4), \
+        f"ours {float(ours)} ref {float(ref)}"
