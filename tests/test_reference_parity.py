@@ -347,3 +347,22 @@ def test_parallel_loss_matches_reference(setups):
     ref = RefLoss(opt, config)(pred, (mask, gt))
     assert torch.allclose(ours.float(), ref.float(), rtol=1e-4), \
         f"ours {float(ours)} ref {float(ref)}"
+
+
+def test_coco_source_convert_matches_reference():
+    """COCO joint-order -> canonical conversion incl. neck synthesis and
+    visibility flags vs reference config/config.py::COCOSourceConfig.convert."""
+    from config.config import (COCOSourceConfig as RefSrc,
+                               GetConfig as RefGetConfig)
+    from improved_body_parts_amd.config import COCOSourceConfig, GetConfig
+    ref_cfg = RefGetConfig("Canonical")
+    cfg = GetConfig("Canonical")
+    rs = np.random.RandomState(4)
+    joints = rs.rand(5, 17, 3).astype(np.float64) * 100
+    joints[:, :, 2] = rs.randint(0, 4, (5, 17))
+    meta = {"joints": joints.copy(), "image": "x.jpg"}
+    ours = COCOSourceConfig("d.h5").convert(dict(meta), cfg)
+    theirs = RefSrc("d.h5").convert({"joints": joints.copy(), "image": "x.jpg"},
+                                    ref_cfg)
+    np.testing.assert_allclose(np.asarray(ours["joints"]),
+                               np.asarray(theirs["joints"]))
