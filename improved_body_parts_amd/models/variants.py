@@ -2,23 +2,26 @@
 
 Capability parity with the reference's model-ablation files, expressed as
 variant classes over the shared HIP-dispatched blocks instead of whole-file
-forks:
+forks. State-dict layouts match the reference variants KEY-FOR-KEY (verified
+by strict checkpoint loads + forward comparison in
+tests/test_reference_parity.py):
 
   * ``PoseNetFinal``      — reference models/posenet_final.py (+
-    layers_transposed_final.py): identity-mapping hourglass (two residual
-    refinements after each upsample, residual-add THEN LeakyReLU), simpler
-    residual-chain backbone, channel-compress 1x1 + two 3x3 before regression.
-  * ``PoseNetAttention``  — reference models/posenet2.py: SE channel attention
-    applied to every hourglass scale on every stack, heads regressed on the
-    UNCOMPRESSED per-scale channels.
-  * ``PoseNetLight``      — reference models/posenet3.py: plain conv stem, a
-    single 3x3 Conv before each regression head, heads on uncompressed
-    channels.
-  * ``PoseNetIndependent``— reference models/posenet_independent.py: no
-    cross-stack residual feature cache (each stack sees only the merged input).
+    layers_transposed_final.py): identity-mapping hourglass built from plain
+    Conv blocks (two refinements after each upsample, residual-add THEN
+    LeakyReLU), residual-chain backbone, SE attention on every scale of every
+    stack, channel-compress 1x1 + two 3x3 before regression.
+  * ``PoseNetAttention``  — reference models/posenet2.py: SE attention on all
+    scales/stacks, heads + merges on the UNCOMPRESSED per-scale channels,
+    bn-free merge convs.
+  * ``PoseNetLight``      — reference models/posenet3.py: plain conv stem, SE
+    attention on all scales/stacks, a single 3x3 Conv before each head, heads
+    + merges on uncompressed channels.
+  * ``PoseNetIndependent``— reference models/posenet_independent.py: built on
+    the CLASSIC hourglass blocks (models/layers.py: plain up1+up2 merge, no
+    post-upsample refine), plain conv stem, no cross-stack feature cache.
   * ``AEPoseNet``         — reference models/ae_pose.py: single-scale
-    Associative-Embedding-style stacked hourglass (output list of [nstack]
-    tensors, one scale each).
+    Associative-Embedding-style stacked hourglass.
 
 All accept ``(nstack, inp_dim, oup_dim, bn, increase)`` like ``PoseNet`` and
 run through the same ops layer (HIP kernels on MI355X, eager on CPU).
@@ -61,11 +64,16 @@ class BackboneFinal(nn.Module):
 
 
 class HourglassFinal(nn.Module):
-    """Identity-mapping hourglass: two residual refinements after each
-    upsample, the residual add comes BEFORE the LeakyReLU
-    (reference layers_transposed_final.py:129-189). Returns 5 scales."""
+    """Identity-mapping hourglass from plain Conv blocks: two refinements
+    after each upsample, residual add BEFORE the LeakyReLU
+    (reference layers_transposed_final.py:110-189). Returns 5 scales.
 
-    def __init__(self, depth, nFeat, increase=128, bn=False, resBlock=Residual):
+    Slot layout per depth (matches the reference's ModuleList indices):
+      0 skip Conv (relu=False) | 1 down Conv (+increase) | 2 pre-upsample Conv
+      (-increase) | 3, 4 refine Convs (4 relu=False) | 5 LeakyReLU |
+      6 innermost Conv (deepest depth only)."""
+
+    def __init__(self, depth, nFeat, increase=128, bn=False, convBlock=Conv):
         super().__init__()
         self.depth = depth
         hg = []
@@ -73,15 +81,15 @@ class HourglassFinal(nn.Module):
             c0 = nFeat + increase * d
             c1 = nFeat + increase * (d + 1)
             mods = [
-                resBlock(c0, c0, bn=bn, relu=False),  # 0: skip (pre-activation)
-                resBlock(c0, c1, bn=bn),              # 1: after downsample
-                resBlock(c1, c0, bn=bn),              # 2: before upsample
-                resBlock(c0, c0, bn=bn),              # 3: refine 1 after upsample
-                resBlock(c0, c0, bn=bn, relu=False),  # 4: refine 2 (no act)
-                nn.LeakyReLU(negative_slope=LEAKY_SLOPE, inplace=True),  # 5
+                convBlock(c0, c0, bn=bn, relu=False),
+                convBlock(c0, c1, bn=bn),
+                convBlock(c1, c0, bn=bn),
+                convBlock(c0, c0, bn=bn),
+                convBlock(c0, c0, bn=bn, relu=False),
+                nn.LeakyReLU(negative_slope=LEAKY_SLOPE, inplace=True),
             ]
             if d == depth - 1:
-                mods.append(resBlock(c1, c1, bn=bn))  # 6: innermost
+                mods.append(convBlock(c1, c1, bn=bn))
             hg.append(nn.ModuleList(mods))
         self.hg = nn.ModuleList(hg)
         self.downsample = nn.MaxPool2d(2, 2)
@@ -126,72 +134,10 @@ class FeaturesCompress(nn.Module):
         return [self.before_regress[i](fms[i]) for i in range(5)]
 
 
-class PoseNetFinal(PoseNet):
-    """Reference models/posenet_final.py: final IMHN ablation."""
-
-    def __init__(self, nstack, inp_dim, oup_dim, bn=False, increase=128,
-                 init_weights=True, **kwargs):
-        super().__init__(nstack, inp_dim, oup_dim, bn=bn, increase=increase,
-                         init_weights=False, **kwargs)
-        self.pre = BackboneFinal(nFeat=inp_dim)
-        self.hourglass = nn.ModuleList([
-            HourglassFinal(4, inp_dim, increase, bn=bn) for _ in range(nstack)])
-        self.features = nn.ModuleList([
-            FeaturesCompress(inp_dim, increase=increase, bn=bn)
-            for _ in range(nstack)])
-        if init_weights:
-            self._initialize_weights()
-
-
-# ---------------------------------------------------------------------------
-# attention variant (reference posenet2.py)
-# ---------------------------------------------------------------------------
-
-class FeaturesWide(nn.Module):
-    """Two 3x3 Convs per scale on the uncompressed channels
-    (reference posenet2.py Features)."""
-
-    def __init__(self, inp_dim, increase=128, bn=False):
-        super().__init__()
-        self.before_regress = nn.ModuleList([
-            nn.Sequential(
-                Conv(inp_dim + i * increase, inp_dim + i * increase, 3, bn=bn),
-                Conv(inp_dim + i * increase, inp_dim + i * increase, 3, bn=bn),
-            ) for i in range(5)
-        ])
-
-    def forward(self, fms):
-        assert len(fms) == 5
-        return [self.before_regress[i](fms[i]) for i in range(5)]
-
-
-class PoseNetAttention(PoseNet):
-    """Reference models/posenet2.py: SE attention on every scale of every
-    stack; heads + merges on the uncompressed per-scale channels."""
-
-    def __init__(self, nstack, inp_dim, oup_dim, bn=False, increase=128,
-                 init_weights=True, **kwargs):
-        super().__init__(nstack, inp_dim, oup_dim, bn=bn, increase=increase,
-                         init_weights=False, **kwargs)
-        self.features = nn.ModuleList([
-            FeaturesWide(inp_dim, increase=increase, bn=bn)
-            for _ in range(nstack)])
-        self.channel_attention = nn.ModuleList([
-            nn.ModuleList([SELayer(inp_dim + j * increase) for j in range(5)])
-            for _ in range(nstack)])
-        self.outs = nn.ModuleList([
-            nn.ModuleList([Conv(inp_dim + j * increase, oup_dim, 1, relu=False,
-                                bn=False) for j in range(5)])
-            for _ in range(nstack)])
-        self.merge_features = nn.ModuleList([
-            nn.ModuleList([Merge(inp_dim + j * increase, inp_dim + j * increase,
-                                 bn=bn) for j in range(5)])
-            for _ in range(nstack - 1)])
-        self.merge_preds = nn.ModuleList([
-            nn.ModuleList([Merge(oup_dim, inp_dim + j * increase, bn=bn)
-                           for j in range(5)]) for _ in range(nstack - 1)])
-        if init_weights:
-            self._initialize_weights()
+class _AttentionStacksMixin:
+    """Shared forward for the variants that SE-attend every hourglass scale
+    before regression (reference posenet2.py/posenet3.py/posenet_final.py
+    forward bodies are identical in structure)."""
 
     def forward(self, imgs):
         x = imgs.permute(0, 3, 1, 2)
@@ -221,6 +167,78 @@ class PoseNetAttention(PoseNet):
         return pred
 
 
+class PoseNetFinal(_AttentionStacksMixin, PoseNet):
+    """Reference models/posenet_final.py: final IMHN ablation."""
+
+    def __init__(self, nstack, inp_dim, oup_dim, bn=False, increase=128,
+                 init_weights=True, **kwargs):
+        super().__init__(nstack, inp_dim, oup_dim, bn=bn, increase=increase,
+                         init_weights=False, **kwargs)
+        self.pre = BackboneFinal(nFeat=inp_dim)
+        self.hourglass = nn.ModuleList([
+            HourglassFinal(4, inp_dim, increase, bn=bn) for _ in range(nstack)])
+        self.features = nn.ModuleList([
+            FeaturesCompress(inp_dim, increase=increase, bn=bn)
+            for _ in range(nstack)])
+        self.channel_attention = nn.ModuleList([
+            nn.ModuleList([SELayer(inp_dim + j * increase) for j in range(5)])
+            for _ in range(nstack)])
+        if init_weights:
+            self._initialize_weights()
+
+
+# ---------------------------------------------------------------------------
+# attention variant (reference posenet2.py)
+# ---------------------------------------------------------------------------
+
+class FeaturesWide(nn.Module):
+    """Two 3x3 Convs per scale on the uncompressed channels
+    (reference posenet2.py Features)."""
+
+    def __init__(self, inp_dim, increase=128, bn=False):
+        super().__init__()
+        self.before_regress = nn.ModuleList([
+            nn.Sequential(
+                Conv(inp_dim + i * increase, inp_dim + i * increase, 3, bn=bn),
+                Conv(inp_dim + i * increase, inp_dim + i * increase, 3, bn=bn),
+            ) for i in range(5)
+        ])
+
+    def forward(self, fms):
+        assert len(fms) == 5
+        return [self.before_regress[i](fms[i]) for i in range(5)]
+
+
+class PoseNetAttention(_AttentionStacksMixin, PoseNet):
+    """Reference models/posenet2.py: SE attention on every scale of every
+    stack; heads + merges on the uncompressed per-scale channels; the merge
+    convs carry a bias and no BN (reference posenet2.py:18)."""
+
+    def __init__(self, nstack, inp_dim, oup_dim, bn=False, increase=128,
+                 init_weights=True, **kwargs):
+        super().__init__(nstack, inp_dim, oup_dim, bn=bn, increase=increase,
+                         init_weights=False, **kwargs)
+        self.features = nn.ModuleList([
+            FeaturesWide(inp_dim, increase=increase, bn=bn)
+            for _ in range(nstack)])
+        self.channel_attention = nn.ModuleList([
+            nn.ModuleList([SELayer(inp_dim + j * increase) for j in range(5)])
+            for _ in range(nstack)])
+        self.outs = nn.ModuleList([
+            nn.ModuleList([Conv(inp_dim + j * increase, oup_dim, 1, relu=False,
+                                bn=False) for j in range(5)])
+            for _ in range(nstack)])
+        self.merge_features = nn.ModuleList([
+            nn.ModuleList([Merge(inp_dim + j * increase, inp_dim + j * increase,
+                                 bn=False) for j in range(5)])
+            for _ in range(nstack - 1)])
+        self.merge_preds = nn.ModuleList([
+            nn.ModuleList([Merge(oup_dim, inp_dim + j * increase, bn=False)
+                           for j in range(5)]) for _ in range(nstack - 1)])
+        if init_weights:
+            self._initialize_weights()
+
+
 # ---------------------------------------------------------------------------
 # light variant (reference posenet3.py)
 # ---------------------------------------------------------------------------
@@ -241,9 +259,10 @@ class FeaturesLight(nn.Module):
         return [self.before_regress[i](fms[i]) for i in range(5)]
 
 
-class PoseNetLight(PoseNet):
-    """Reference models/posenet3.py: plain conv stem, single pre-regress conv,
-    heads + merges on the uncompressed channels."""
+class PoseNetLight(_AttentionStacksMixin, PoseNet):
+    """Reference models/posenet3.py: plain conv stem, SE attention on all
+    scales/stacks, single pre-regress conv, heads + merges on the uncompressed
+    channels (merges keep BN, reference posenet3.py:21)."""
 
     def __init__(self, nstack, inp_dim, oup_dim, bn=False, increase=128,
                  init_weights=True, **kwargs):
@@ -259,6 +278,9 @@ class PoseNetLight(PoseNet):
         self.features = nn.ModuleList([
             FeaturesLight(inp_dim, increase=increase, bn=bn)
             for _ in range(nstack)])
+        self.channel_attention = nn.ModuleList([
+            nn.ModuleList([SELayer(inp_dim + j * increase) for j in range(5)])
+            for _ in range(nstack)])
         self.outs = nn.ModuleList([
             nn.ModuleList([Conv(inp_dim + j * increase, oup_dim, 1, relu=False,
                                 bn=False) for j in range(5)])
@@ -273,39 +295,89 @@ class PoseNetLight(PoseNet):
         if init_weights:
             self._initialize_weights()
 
-    def forward(self, imgs):
-        x = imgs.permute(0, 3, 1, 2)
-        if x.is_cuda:
-            x = x.contiguous(memory_format=torch.channels_last)
-        x = self.pre(x)
-        pred = []
-        features_cache = [None] * 5
-        for i in range(self.nstack):
-            hourglass_feature = self.hourglass[i](x)
-            if i > 0:
-                hourglass_feature = [hourglass_feature[s] + features_cache[s]
-                                     for s in range(5)]
-            features_instack = self.features[i](hourglass_feature)
-            preds_instack = []
-            for j in range(5):
-                preds_instack.append(self.outs[i][j](features_instack[j]))
-                if i != self.nstack - 1:
-                    merged = (self.merge_preds[i][j](preds_instack[j])
-                              + self.merge_features[i][j](features_instack[j]))
-                    if j == 0:
-                        x = x + merged
-                    features_cache[j] = merged
-            pred.append(preds_instack)
-        return pred
-
 
 # ---------------------------------------------------------------------------
-# independent-stack variant (reference posenet_independent.py)
+# independent-stack variant (reference posenet_independent.py + models/layers.py)
 # ---------------------------------------------------------------------------
 
-class PoseNetIndependent(PoseNet):
-    """Reference models/posenet_independent.py: stacks do NOT share the
-    residual feature cache — only the scale-0 merge feeds forward."""
+class HourglassClassic(nn.Module):
+    """The original (Associative-Embedding-style) hourglass: 3 Conv slots per
+    depth + innermost, plain up1 + up2 merge with NO post-upsample refinement
+    (reference models/layers.py:81-169). Returns 5 scales."""
+
+    def __init__(self, depth, nFeat, increase=128, bn=False, convBlock=Conv):
+        super().__init__()
+        self.depth = depth
+        hg = []
+        for d in range(depth):
+            c0 = nFeat + increase * d
+            c1 = nFeat + increase * (d + 1)
+            mods = [
+                convBlock(c0, c0, bn=bn),
+                convBlock(c0, c1, bn=bn),
+                convBlock(c1, c0, bn=bn),
+            ]
+            if d == depth - 1:
+                mods.append(convBlock(c1, c1, bn=bn))
+            hg.append(nn.ModuleList(mods))
+        self.hg = nn.ModuleList(hg)
+        self.downsample = nn.MaxPool2d(2, 2)
+        self.upsample = nn.Upsample(scale_factor=2, mode="nearest")
+
+    def _forward(self, d, x, up_fms):
+        up1 = self.hg[d][0](x)
+        low1 = ops.maxpool2x2(x)
+        low1 = self.hg[d][1](low1)
+        if d == self.depth - 1:
+            low2 = self.hg[d][3](low1)
+        else:
+            low2 = self._forward(d + 1, low1, up_fms)
+        low3 = self.hg[d][2](low2)
+        up_fms.append(low2)
+        up2 = ops.upsample2x_nearest(low3)
+        return up1 + up2
+
+    def forward(self, x):
+        up_fms = []
+        top = self._forward(0, x, up_fms)
+        return [top] + up_fms[::-1]
+
+
+class PoseNetIndependent(nn.Module):
+    """Reference models/posenet_independent.py: classic hourglass blocks,
+    plain conv stem, two uncompressed pre-regress convs, bias-carrying merge
+    convs, and NO cross-stack residual feature cache."""
+
+    def __init__(self, nstack, inp_dim, oup_dim, bn=False, increase=128,
+                 init_weights=True, **kwargs):
+        super().__init__()
+        self.nstack = nstack
+        self.pre = nn.Sequential(
+            Conv(3, 64, 7, 2, bn=bn),
+            Conv(64, 128, bn=bn),
+            nn.MaxPool2d(2, 2),
+            Conv(128, 128, bn=bn),
+            Conv(128, inp_dim, bn=bn),
+        )
+        self.hourglass = nn.ModuleList([
+            HourglassClassic(4, inp_dim, increase, bn=bn)
+            for _ in range(nstack)])
+        self.features = nn.ModuleList([
+            FeaturesWide(inp_dim, increase=increase, bn=bn)
+            for _ in range(nstack)])
+        self.outs = nn.ModuleList([
+            nn.ModuleList([Conv(inp_dim + j * increase, oup_dim, 1, relu=False,
+                                bn=False) for j in range(5)])
+            for _ in range(nstack)])
+        self.merge_features = nn.ModuleList([
+            nn.ModuleList([Merge(inp_dim + j * increase, inp_dim + j * increase,
+                                 bn=False) for j in range(5)])
+            for _ in range(nstack - 1)])
+        self.merge_preds = nn.ModuleList([
+            nn.ModuleList([Merge(oup_dim, inp_dim + j * increase, bn=False)
+                           for j in range(5)]) for _ in range(nstack - 1)])
+        if init_weights:
+            PoseNet._initialize_weights(self)
 
     def forward(self, imgs):
         x = imgs.permute(0, 3, 1, 2)

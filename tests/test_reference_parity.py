@@ -366,3 +366,31 @@ def test_coco_source_convert_matches_reference():
                                     ref_cfg)
     np.testing.assert_allclose(np.asarray(ours["joints"]),
                                np.asarray(theirs["joints"]))
+
+
+@pytest.mark.parametrize("refmod,ourname", [
+    ("models.posenet_final", "PoseNetFinal"),
+    ("models.posenet2", "PoseNetAttention"),
+    ("models.posenet3", "PoseNetLight"),
+    ("models.posenet_independent", "PoseNetIndependent"),
+])
+def test_variant_checkpoint_and_forward_match_reference(refmod, ourname):
+    """Every model-variant family loads the corresponding reference variant's
+    state_dict STRICT and reproduces its forward outputs."""
+    import importlib
+    _import_reference_evaluate()
+    import improved_body_parts_amd.models as M
+    rm = importlib.import_module(refmod)
+    torch.manual_seed(0)
+    ref = rm.PoseNet(2, 256, 50, bn=True, increase=32)
+    ours = getattr(M, ourname)(2, 256, 50, bn=True, increase=32)
+    ours.load_state_dict(ref.state_dict())  # strict
+    ref.eval()
+    ours.eval()
+    x = torch.rand(1, 64, 64, 3)
+    with torch.no_grad():
+        a = ref(x)
+        b = ours(x)
+    for s in range(5):
+        rel = float((b[1][s] - a[1][s]).norm() / (a[1][s].norm() + 1e-12))
+        assert rel < 1e-5, f"{ourname} scale {s}: rel {rel}"
