@@ -399,37 +399,99 @@ class PoseNetIndependent(nn.Module):
 
 
 # ---------------------------------------------------------------------------
-# Associative-Embedding-style single-scale stack (reference ae_pose.py)
+# Associative-Embedding-style single-scale stack (reference ae_pose.py +
+# ae_layer.py — "Copied from Associative Embedding pytorch project")
 # ---------------------------------------------------------------------------
 
+class AEConv(nn.Module):
+    """AE-style conv block: conv ALWAYS carries a bias, plain ReLU (not
+    leaky), and the activation comes BEFORE the BN (reference ae_layer.py
+    Conv)."""
+
+    def __init__(self, inp_dim, out_dim, kernel_size=3, stride=1, bn=False,
+                 relu=True):
+        super().__init__()
+        self.inp_dim = inp_dim
+        self.conv = nn.Conv2d(inp_dim, out_dim, kernel_size, stride,
+                              padding=(kernel_size - 1) // 2, bias=True)
+        self.relu = nn.ReLU() if relu else None
+        self.bn = nn.BatchNorm2d(out_dim) if bn else None
+
+    def forward(self, x):
+        assert x.size(1) == self.inp_dim
+        x = self.conv(x)
+        if self.relu is not None:
+            x = self.relu(x)
+        if self.bn is not None:
+            x = self.bn(x)
+        return x
+
+
+class AEHourglass(nn.Module):
+    """Recursive attribute-style hourglass, single-scale output
+    (reference ae_layer.py Hourglass): up1 + upsample(low3(low2(low1(pool))))."""
+
+    def __init__(self, n, f, bn=None, increase=128):
+        super().__init__()
+        nf = f + increase
+        self.up1 = AEConv(f, f, 3, bn=bn)
+        self.pool1 = nn.MaxPool2d(2, 2)
+        self.low1 = AEConv(f, nf, 3, bn=bn)
+        if n > 1:
+            self.low2 = AEHourglass(n - 1, nf, bn=bn, increase=increase)
+        else:
+            self.low2 = AEConv(nf, nf, 3, bn=bn)
+        self.low3 = AEConv(nf, f, 3)
+        self.up2 = nn.UpsamplingNearest2d(scale_factor=2)
+
+    def forward(self, x):
+        up1 = self.up1(x)
+        low1 = self.low1(ops.maxpool2x2(x))
+        low2 = self.low2(low1)
+        low3 = self.low3(low2)
+        return up1 + ops.upsample2x_nearest(low3)
+
+
+class AEMerge(nn.Module):
+    """1x1 channel adapter (reference ae_pose.py Merge)."""
+
+    def __init__(self, x_dim, y_dim):
+        super().__init__()
+        self.conv = AEConv(x_dim, y_dim, 1, relu=False, bn=False)
+
+    def forward(self, x):
+        return self.conv(x)
+
+
 class AEPoseNet(nn.Module):
-    """Single-scale stacked hourglass (reference models/ae_pose.py:21-77):
-    conv stem, per-stack [hourglass -> feature conv -> 1x1 head], inter-stack
-    merge at one scale. Output: list of [nstack] (N, oup_dim, H/4, W/4)."""
+    """Single-scale stacked hourglass (reference models/ae_pose.py:20-58):
+    AE conv stem, per-stack [hourglass -> 2 feature convs -> 1x1 head],
+    inter-stack merge at one scale. Output: [nstack] x (N, oup_dim, H/4, W/4).
+    State-dict layout matches the reference exactly."""
 
     def __init__(self, nstack, inp_dim, oup_dim, bn=False, increase=128,
                  init_weights=True, **kwargs):
         super().__init__()
         self.nstack = nstack
         self.pre = nn.Sequential(
-            Conv(3, 64, 7, 2, bn=bn),
-            Conv(64, 128, bn=bn),
+            AEConv(3, 64, 7, 2, bn=bn),
+            AEConv(64, 128, bn=bn),
             nn.MaxPool2d(2, 2),
-            Conv(128, 128, bn=bn),
-            Conv(128, inp_dim, bn=bn),
+            AEConv(128, 128, bn=bn),
+            AEConv(128, inp_dim, bn=bn),
         )
         self.features = nn.ModuleList([
             nn.Sequential(
-                Hourglass(4, inp_dim, increase, bn=bn),
-                _TakeScale0(),
-                Conv(inp_dim, inp_dim, 3, bn=False),
-                Conv(inp_dim, inp_dim, 3, bn=False),
+                AEHourglass(4, inp_dim, bn, increase),
+                AEConv(inp_dim, inp_dim, 3, bn=False),
+                AEConv(inp_dim, inp_dim, 3, bn=False),
             ) for _ in range(nstack)])
-        self.outs = nn.ModuleList([Conv(inp_dim, oup_dim, 1, relu=False, bn=False)
-                                   for _ in range(nstack)])
-        self.merge_features = nn.ModuleList([Merge(inp_dim, inp_dim)
+        self.outs = nn.ModuleList([
+            AEConv(inp_dim, oup_dim, 1, relu=False, bn=False)
+            for _ in range(nstack)])
+        self.merge_features = nn.ModuleList([AEMerge(inp_dim, inp_dim)
                                              for _ in range(nstack - 1)])
-        self.merge_preds = nn.ModuleList([Merge(oup_dim, inp_dim)
+        self.merge_preds = nn.ModuleList([AEMerge(oup_dim, inp_dim)
                                           for _ in range(nstack - 1)])
         if init_weights:
             PoseNet._initialize_weights(self)
@@ -444,16 +506,8 @@ class AEPoseNet(nn.Module):
             feature = self.features[i](x)
             preds.append(self.outs[i](feature))
             if i != self.nstack - 1:
-                x = x + self.merge_preds[i](preds[-1]) \
-                    + self.merge_features[i](feature)
+                x = x + self.merge_preds[i](preds[-1])                     + self.merge_features[i](feature)
         return preds
-
-
-class _TakeScale0(nn.Module):
-    """Adapter: our Hourglass returns 5 scales; the AE variant uses scale 0."""
-
-    def forward(self, fms):
-        return fms[0] if isinstance(fms, (list, tuple)) else fms
 
 
 VARIANTS = {

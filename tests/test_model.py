@@ -140,9 +140,9 @@ def test_ae_variant_forward_backward():
     net = build_posenet("ae", nstack=2, inp_dim=64, oup_dim=50, bn=True,
                         increase=32)
     out = net(torch.rand(1, 128, 128, 3))
-    assert len(out) == 2
-    assert out[0].shape == (1, 50, 32, 32)
-    out[-1].sum().backward()
+    assert len(out) == 2 and len(out[0]) == 1
+    assert out[0][0].shape == (1, 50, 32, 32)
+    out[-1][0].sum().backward()
 
 
 def test_variant_registry_rejects_unknown():

@@ -394,3 +394,25 @@ def test_variant_checkpoint_and_forward_match_reference(refmod, ourname):
     for s in range(5):
         rel = float((b[1][s] - a[1][s]).norm() / (a[1][s].norm() + 1e-12))
         assert rel < 1e-5, f"{ourname} scale {s}: rel {rel}"
+
+
+def test_ae_variant_checkpoint_and_forward_match_reference():
+    """AEPoseNet loads the reference ae_pose state_dict strict and reproduces
+    its forward (single-scale [nstack][1] output nesting included)."""
+    import importlib
+    _import_reference_evaluate()
+    import improved_body_parts_amd.models as M
+    rm = importlib.import_module("models.ae_pose")
+    torch.manual_seed(0)
+    ref = rm.PoseNet(2, 256, 50, bn=True)
+    ours = M.AEPoseNet(2, 256, 50, bn=True)
+    ours.load_state_dict(ref.state_dict())
+    ref.eval()
+    ours.eval()
+    x = torch.rand(1, 64, 64, 3)
+    with torch.no_grad():
+        a = ref(x)
+        b = ours(x)
+    for i in range(2):
+        rel = float((b[i][0] - a[i][0]).norm() / (a[i][0].norm() + 1e-12))
+        assert rel < 1e-5, f"stack {i}: {rel}"
