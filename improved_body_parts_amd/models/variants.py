@@ -504,9 +504,13 @@ class AEPoseNet(nn.Module):
         preds = []
         for i in range(self.nstack):
             feature = self.features[i](x)
-            preds.append(self.outs[i](feature))
+            pred = self.outs[i](feature)
             if i != self.nstack - 1:
-                x = x + self.merge_preds[i](preds[-1])                     + self.merge_features[i](feature)
+                x = x + self.merge_preds[i](pred) \
+                    + self.merge_features[i](feature)
+            # [nstack][1] nesting mirrors the reference (ae_pose.py:49-57) so
+            # the Network/loss wrappers see the usual [stack][scale] shape
+            preds.append([pred])
         return preds
 
 
