@@ -26,6 +26,7 @@ class TrainingOpt:
         self.hdf5_train_data = "./data/dataset/coco/link2coco2017/coco_train_dataset512.h5"
         self.hdf5_val_data = "./data/dataset/coco/link2coco2017/coco_val_dataset512.h5"
         self.nstack = 4                # number of stacked hourglasses
+        self.model_variant = "imhn"    # imhn | final | attention | light | independent | ae
         self.hourglass_inp_dim = 256   # channels entering each hourglass
         self.increase = 128            # channel increase per down-sampling inside the hourglass
         self.nstack_weight = [1, 1, 1, 1]

@@ -125,14 +125,26 @@ class PoseNet(nn.Module):
                 m.bias.data.zero_()
 
 
+def _build(opt, config, bn, init_weights):
+    """Model factory honouring opt.model_variant (reference: one Network
+    wrapper per variant file; here one wrapper over the variant registry)."""
+    variant = getattr(opt, "model_variant", "imhn")
+    if variant in (None, "imhn"):
+        return PoseNet(opt.nstack, opt.hourglass_inp_dim, config.num_layers,
+                       bn=bn, increase=opt.increase, init_weights=init_weights)
+    from .variants import build_posenet
+    return build_posenet(variant, opt.nstack, opt.hourglass_inp_dim,
+                         config.num_layers, bn=bn, increase=opt.increase,
+                         init_weights=init_weights)
+
+
 class Network(nn.Module):
     """Model + loss fused in one module so every rank/replica computes its own loss
     (reference models/posenet.py:142-173)."""
 
     def __init__(self, opt, config, bn=False, dist=False, swa=False):
         super().__init__()
-        self.posenet = PoseNet(opt.nstack, opt.hourglass_inp_dim, config.num_layers,
-                               bn=bn, increase=opt.increase)
+        self.posenet = _build(opt, config, bn, init_weights=True)
         self.criterion = MultiTaskLoss(opt, config) if dist else MultiTaskLossParallel(opt, config)
         self.swa = swa
 
@@ -153,8 +165,7 @@ class NetworkEval(nn.Module):
 
     def __init__(self, opt, config, bn=False):
         super().__init__()
-        self.posenet = PoseNet(opt.nstack, opt.hourglass_inp_dim, config.num_layers,
-                               bn=bn, init_weights=False, increase=opt.increase)
+        self.posenet = _build(opt, config, bn, init_weights=False)
 
     def forward(self, inp_imgs):
         if self.training:

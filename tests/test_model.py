@@ -149,3 +149,18 @@ def test_variant_registry_rejects_unknown():
     from improved_body_parts_amd.models import build_posenet
     with pytest.raises(ValueError):
         build_posenet("nope", 1, 64, 50)
+
+
+def test_network_with_variant():
+    """Network honours opt.model_variant (driver-level variant selection)."""
+    from improved_body_parts_amd.config import CanonicalConfig, TrainingOpt
+    from improved_body_parts_amd.models import Network
+    from improved_body_parts_amd.models.variants import PoseNetLight
+    cfg = CanonicalConfig(128, 128, 4)
+    opt = TrainingOpt(nstack=2, hourglass_inp_dim=64, increase=32, batch_size=1,
+                      nstack_weight=[1, 1], model_variant="light")
+    net = Network(opt, cfg, bn=True, dist=True)
+    assert isinstance(net.posenet, PoseNetLight)
+    loss = net((torch.rand(1, 128, 128, 3), torch.ones(1, 1, 32, 32),
+                torch.rand(1, 50, 32, 32)))
+    assert torch.isfinite(loss)
