@@ -33,6 +33,8 @@ def main():
     ap.add_argument("--config", default="Canonical",
                     help="Canonical | Canonical384 | Canonical768 | DenseSkeleton")
     ap.add_argument("--nstack", type=int, default=4)
+    ap.add_argument("--variant", default="imhn",
+                    help="imhn | final | attention | light | independent | ae")
     ap.add_argument("--batch", type=int, default=16, help="per-GPU batch size")
     ap.add_argument("--epochs", type=int, default=60)
     ap.add_argument("--lr", type=float, default=2.5e-5, help="base LR per GPU")
@@ -62,7 +64,7 @@ def main():
 
     config = GetConfig(args.config)
     opt = TrainingOpt(nstack=args.nstack, batch_size=args.batch,
-                      learning_rate=args.lr,
+                      learning_rate=args.lr, model_variant=args.variant,
                       nstack_weight=[1] * args.nstack)
 
     if args.data == "synthetic":
