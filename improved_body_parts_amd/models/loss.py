@@ -49,15 +49,19 @@ class MultiTaskLoss(nn.Module):
         self.gamma = gamma
 
     def forward(self, pred_tuple, target_tuple):
-        """pred_tuple: [nstack][5] tensors (N,C,Hs,Ws); target_tuple: (mask_miss (N,1,H,W),
-        heatmaps (N,C,H,W)). Returns a scalar loss (averaged over batch)."""
+        """pred_tuple: [nstack][n_scales] tensors (N,C,Hs,Ws); target_tuple:
+        (mask_miss (N,1,H,W), heatmaps (N,C,H,W)). Returns a scalar loss
+        (averaged over batch). n_scales is 5 for the IMHN families and 1 for
+        the single-scale AE variant."""
         nstack = len(pred_tuple)
+        n_scales = len(pred_tuple[0])
         batch = pred_tuple[0][0].shape[0]
+        sw = self.scale_weight[:n_scales] if n_scales > 1 else [1.0]
         loss_scales = []
-        for i in range(5):
+        for i in range(n_scales):
             pred = torch.stack([pred_tuple[j][i] for j in range(nstack)], dim=0)
-            loss_scales.append(self._loss_per_scale(pred, target_tuple) * self.scale_weight[i])
-        return sum(loss_scales) / sum(self.scale_weight) / batch
+            loss_scales.append(self._loss_per_scale(pred, target_tuple) * sw[i])
+        return sum(loss_scales) / sum(sw) / batch
 
     def _loss_per_scale(self, pred, target):
         size = pred.shape[-2:]
@@ -113,12 +117,13 @@ class MultiTaskLossParallel(MultiTaskLoss):
 
     def forward(self, pred_tuple, target_tuple):
         nstack = len(pred_tuple)
+        n_scales = len(pred_tuple[0])
+        sw = self.scale_weight[:n_scales] if n_scales > 1 else [1.0]
         loss_scales = []
-        for i in range(5):
+        for i in range(n_scales):
             pred = torch.stack([pred_tuple[j][i] for j in range(nstack)], dim=0)
-            loss_scales.append(self._loss_per_scale(pred, target_tuple)
-                               * self.scale_weight[i])
-        return sum(loss_scales) / sum(self.scale_weight)
+            loss_scales.append(self._loss_per_scale(pred, target_tuple) * sw[i])
+        return sum(loss_scales) / sum(sw)
 
     def _loss_per_scale(self, pred, target):
         size = pred.shape[-2:]
