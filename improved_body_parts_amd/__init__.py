@@ -20,4 +20,13 @@ Layout:
   engine/    training / evaluation drivers, checkpointing
   utils/     padding, NMS, meters, logging
 """
+import os as _os
+
+# The few library-conv fallbacks (7x7 Cin=3 stem, wgrad) depend on MIOpen's
+# solver choice; its default DYNAMIC_HYBRID find mode picked the 2.5 ms
+# naive_conv kernel for the stem on a cold box (rocprof, profiles/README.md)
+# where NORMAL find picks the 38 us igemm. Pay the one-time find cost during
+# warmup instead of losing ~20% of the steady-state step to naive kernels.
+_os.environ.setdefault("MIOPEN_FIND_MODE", "1")
+
 __version__ = "0.1.0"
