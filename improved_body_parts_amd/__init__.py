@@ -23,10 +23,10 @@ Layout:
 import os as _os
 
 # The few library-conv fallbacks (7x7 Cin=3 stem, wgrad) depend on MIOpen's
-# solver choice; its default DYNAMIC_HYBRID find mode picked the 2.5 ms
-# naive_conv kernel for the stem on a cold box (rocprof, profiles/README.md)
-# where NORMAL find picks the 38 us igemm. Pay the one-time find cost during
-# warmup instead of losing ~20% of the steady-state step to naive kernels.
+# solver choice. NORMAL find benchmarks every applicable solver on first use
+# (cost sits in warmup; rocprof shows its naive-kernel evaluations there) and
+# then pins the best — steady-state solver choice stays deterministic across
+# fresh boxes instead of depending on the find-db the box happens to have.
 _os.environ.setdefault("MIOPEN_FIND_MODE", "1")
 
 __version__ = "0.1.0"
