@@ -20,13 +20,4 @@ Layout:
   engine/    training / evaluation drivers, checkpointing
   utils/     padding, NMS, meters, logging
 """
-import os as _os
-
-# The few library-conv fallbacks (7x7 Cin=3 stem, wgrad) depend on MIOpen's
-# solver choice. NORMAL find benchmarks every applicable solver on first use
-# (cost sits in warmup; rocprof shows its naive-kernel evaluations there) and
-# then pins the best — steady-state solver choice stays deterministic across
-# fresh boxes instead of depending on the find-db the box happens to have.
-_os.environ.setdefault("MIOPEN_FIND_MODE", "1")
-
 __version__ = "0.1.0"
