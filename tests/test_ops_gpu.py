@@ -249,7 +249,7 @@ def test_focal_l2_gpu_vs_cpu(gamma):
     pr = pred.clone().requires_grad_(True)
     lg = ops.focal_l2_loss(pg, gt.cuda(), mask.cuda(), **kw)
     lr = ops.focal_l2_loss(pr, gt, mask, **kw)
-    _assert_close(lg, lr, 1e-4, 1e-3 * float(lr), "loss")
+    _assert_close(lg, lr, 1e-4, 1e-3 * float(lr.detach()), "loss")
     lg.backward()
     lr.backward()
     _assert_close(pg.grad, pr.grad, 1e-4, 1e-5, "dpred")
