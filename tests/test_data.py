@@ -93,3 +93,33 @@ def test_dataloader_integration(cfg):
     img, mm, hm = next(iter(loader))
     assert img.shape == (3, 128, 128, 3)
     assert hm.shape == (3, 50, 32, 32)
+
+
+def test_transform_image_joint_alignment(cfg):
+    """A bright delta painted at a joint must land at the transformed joint
+    coordinate after the SAME affine warps the image (full random aug)."""
+    import numpy as np
+    from improved_body_parts_amd.data import Transformer, AugmentSelection
+    import random as _random
+    rng = _random.Random(5)
+    t = Transformer(cfg)
+    for trial in range(5):
+        jx, jy = rng.uniform(100, 400), rng.uniform(100, 400)
+        img = np.zeros((512, 512, 3), np.float32)
+        y0, x0 = int(jy), int(jx)
+        img[y0 - 2:y0 + 3, x0 - 2:x0 + 3] = 1.0
+        joints = np.zeros((1, cfg.num_parts, 3), np.float32)
+        joints[:, :, 2] = 2.0
+        joints[0, 0] = (jx, jy, 1.0)
+        meta = {"joints": joints.copy(), "objpos": [jx, jy],
+                "scale_provided": 150.0 / 368}
+        mask = np.ones((512, 512), np.float32)
+        aug = AugmentSelection.random(cfg.transform_params, rng=rng)
+        out_img, _, _, out_meta = t.transform(img, mask.copy(), mask.copy(),
+                                              dict(meta), aug=aug)
+        nx, ny = out_meta["joints"][0, 0, :2]
+        if not (4 <= nx < cfg.width - 4 and 4 <= ny < cfg.height - 4):
+            continue  # joint warped out of frame — nothing to check
+        window = out_img[int(ny) - 4:int(ny) + 5, int(nx) - 4:int(nx) + 5]
+        assert window.max() > 0.2, \
+            f"trial {trial}: no bright pixel near transformed joint ({nx},{ny})"

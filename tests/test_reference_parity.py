@@ -416,3 +416,29 @@ def test_ae_variant_checkpoint_and_forward_match_reference():
     for i in range(2):
         rel = float((b[i][0] - a[i][0]).norm() / (a[i][0].norm() + 1e-12))
         assert rel < 1e-5, f"stack {i}: {rel}"
+
+
+def test_our_checkpoint_loads_into_reference():
+    """Inverse interop direction: a checkpoint written by OUR trainer loads
+    strict into the reference PoseNet."""
+    import tempfile
+    _import_reference_evaluate()
+    from models.posenet import PoseNet as RefPoseNet
+    from improved_body_parts_amd.config import CanonicalConfig, TrainingOpt
+    from improved_body_parts_amd.engine import save_checkpoint
+    from improved_body_parts_amd.engine.optimizer import FusedSGD
+    from improved_body_parts_amd.models import Network
+    cfg = CanonicalConfig(128, 128, 4)
+    opt = TrainingOpt(nstack=2, hourglass_inp_dim=256, batch_size=1,
+                      nstack_weight=[1, 1])
+    net = Network(opt, cfg, bn=True, dist=True)
+    sgd = FusedSGD(net.parameters(), lr=1e-5)
+    with tempfile.TemporaryDirectory() as d:
+        path = save_checkpoint(net, sgd, 1.23, 7, d)
+        ckpt = torch.load(path, map_location="cpu", weights_only=False)
+    assert set(ckpt) == {"weights", "optimizer_weight", "train_loss", "epoch"}
+    ref = RefPoseNet(2, 256, 50, bn=True, increase=128)
+    # our Network prefixes the model under 'posenet.'
+    weights = {k[len("posenet."):]: v for k, v in ckpt["weights"].items()
+               if k.startswith("posenet.")}
+    ref.load_state_dict(weights)  # strict
