@@ -442,3 +442,20 @@ def test_our_checkpoint_loads_into_reference():
     weights = {k[len("posenet."):]: v for k, v in ckpt["weights"].items()
                if k.startswith("posenet.")}
     ref.load_state_dict(weights)  # strict
+
+
+def test_offset_maps_reference_path_is_broken_upstream():
+    """The reference's put_offset cannot run under its own Canonical config:
+    config.offset_layers is 2 (shared offsets) while put_offset asserts a
+    2*num_parts depth (py_data_heatmapper.py:290) — dead code upstream
+    (disabled in the default pipeline, py_data_iterator.py:64). Our
+    implementation keeps the documented shared-offset semantics and is
+    covered by tests/test_heatmapper.py::test_offset_maps; this test pins the
+    upstream defect so the divergence is deliberate."""
+    _stub_ref_modules()
+    from config.config import GetConfig as RefGetConfig
+    from py_cocodata_server.py_data_heatmapper import Heatmapper as RefHeatmapper
+    ref_hm = RefHeatmapper(RefGetConfig("Canonical"))
+    joints = np.zeros((1, 18, 3), np.float32)
+    with pytest.raises(AssertionError):
+        ref_hm.put_offset(joints)
