@@ -96,3 +96,29 @@ def test_slim384_variant():
     assert c.num_layers == 24 + 18 + 2
     assert (c.width, c.height) == (384, 384)
     assert len(c.flip_paf_ord) == 24
+
+
+@pytest.mark.parametrize("name,side,paf", [("Canonical384", 384, 30),
+                                           ("Canonical768", 768, 30),
+                                           ("DenseSkeleton", 512, 49),
+                                           ("Slim384", 384, 24)])
+def test_config_variants_end_to_end_shapes(name, side, paf):
+    """Every config preset drives the GT generator + loss at its own geometry."""
+    import torch
+    from improved_body_parts_amd.config import TrainingOpt
+    from improved_body_parts_amd.data import Heatmapper, sample_people
+    from improved_body_parts_amd.models import MultiTaskLoss
+    cfg = GetConfig(name)
+    assert (cfg.width, cfg.paf_layers) == (side, paf)
+    grid = side // cfg.stride
+    rng = np.random.default_rng(0)
+    people = sample_people(rng, cfg.width, cfg.height, max_people=2)
+    maps = Heatmapper(cfg).create_heatmaps(people, np.ones((grid, grid), np.float32))
+    assert maps.shape == (cfg.num_layers, grid, grid)
+    opt = TrainingOpt(nstack=1, batch_size=1, nstack_weight=[1])
+    crit = MultiTaskLoss(opt, cfg)
+    preds = [[torch.rand(1, cfg.num_layers, grid // 2 ** s, grid // 2 ** s) * 0.1
+              for s in range(5)]]
+    loss = crit(preds, (torch.ones(1, 1, grid, grid),
+                        torch.from_numpy(maps)[None]))
+    assert torch.isfinite(loss)

@@ -223,3 +223,56 @@ def test_predict_multiscale_and_rotation(config):
         heat, paf = predict(img, model, config, p, mp)
     assert heat.shape[:2] == (64, 64)
     assert torch.isfinite(heat).all() and torch.isfinite(paf).all()
+
+
+def test_find_people_invariants_fuzz(config, params):
+    """Structural invariants of the greedy assembly on random synthetic
+    connection graphs: no candidate id is assigned to two people, part counts
+    match assigned slots, every surviving person has >= 2 parts."""
+    p, _ = params
+    rng = np.random.default_rng(0)
+    for trial in range(25):
+        n_per_part = rng.integers(0, 3, config.heat_layers)
+        all_peaks = []
+        gid = 0
+        for part in range(config.heat_layers):
+            sub = []
+            for _ in range(n_per_part[part]):
+                sub.append((float(rng.uniform(0, 127)), float(rng.uniform(0, 127)),
+                            float(rng.uniform(0.2, 1.0)), gid))
+                gid += 1
+            all_peaks.append(sub)
+        connection_all, special_k = [], []
+        for k, (a, b) in enumerate(config.limbs_conn):
+            nA, nB = len(all_peaks[a]), len(all_peaks[b])
+            if nA == 0 or nB == 0:
+                special_k.append(k)
+                connection_all.append([])
+                continue
+            rows = []
+            used_i, used_j = set(), set()
+            order = [(i, j) for i in range(nA) for j in range(nB)]
+            rng.shuffle(order)
+            for i, j in order:
+                if i in used_i or j in used_j:
+                    continue
+                if rng.random() < 0.6:
+                    continue  # drop some candidate connections
+                used_i.add(i)
+                used_j.add(j)
+                rows.append([all_peaks[a][i][3], all_peaks[b][j][3],
+                             float(rng.uniform(0.1, 1.5)), i, j,
+                             float(rng.uniform(3, 120))])
+            connection_all.append(np.asarray(rows, np.float64).reshape(-1, 6))
+        subset, candidate = find_people(connection_all, special_k, all_peaks,
+                                        p, config)
+        seen_ids = set()
+        for s in subset:
+            ids = [int(v) for v in s[:config.heat_layers, 0] if v >= 0]
+            assert len(ids) == len(set(ids)), "duplicate part inside one person"
+            for v in ids:
+                assert v not in seen_ids, "candidate assigned to two people"
+                seen_ids.add(v)
+            assert s[-1][0] >= 2
+            assert len(ids) == int(s[-1][0]), \
+                f"count {s[-1][0]} != assigned {len(ids)}"
