@@ -13,8 +13,13 @@ import pytest
 import torch
 
 REF = "/root/reference"
-pytestmark = pytest.mark.skipif(not os.path.isdir(REF),
-                                reason="reference repo not mounted")
+pytestmark = [
+    pytest.mark.skipif(not os.path.isdir(REF),
+                       reason="reference repo not mounted"),
+    # the reference triggers scipy/torch deprecation warnings on import
+    pytest.mark.filterwarnings("ignore::DeprecationWarning"),
+    pytest.mark.filterwarnings("ignore::UserWarning"),
+]
 
 if os.path.isdir(REF):
     sys.path.insert(0, REF)
