@@ -276,3 +276,25 @@ def test_find_people_invariants_fuzz(config, params):
             assert s[-1][0] >= 2
             assert len(ids) == int(s[-1][0]), \
                 f"count {s[-1][0]} != assigned {len(ids)}"
+
+
+@pytest.mark.gpu
+def test_process_end_to_end_gpu(config):
+    """Full pipeline on GPU: bf16 model ensemble forward (device-resident
+    predict) -> HIP peak/limb kernels -> host assembly."""
+    opt = TrainingOpt(nstack=1, batch_size=1)
+    model = NetworkEval(opt, config, bn=True).cuda().bfloat16()
+    for m in model.modules():
+        if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+            m.float()
+    model.eval()
+    img = np.random.RandomState(1).rand(128, 128, 3).astype(np.float32)
+    p, mp = InferenceParams().as_params_dict()
+    mp = dict(mp)
+    mp["boxsize"] = 128
+    heat, paf = predict(img, model, config, p, mp)
+    assert heat.is_cuda and paf.is_cuda
+    assert heat.shape == (128, 128, config.num_layers - config.paf_layers)
+    assert torch.isfinite(heat).all() and torch.isfinite(paf).all()
+    kps = process(img, model, config, p, mp)
+    assert isinstance(kps, list)
