@@ -4,13 +4,16 @@
 Measures the flagship 4-stage IMHN @512x512 bf16 on synthetic data with
 random-init weights (no network access in this environment):
 
-  * --mode train (default): full training step — synthetic batch -> forward ->
-    focal-L2 loss -> backward (with overlapped RCCL all-reduce when N > 1) ->
-    fused SGD step. Metric: aggregate train_images_per_sec over all ranks
-    (weak scaling: fixed per-GPU batch).
-  * --mode infer: forward-only inference, batch 4, last-stack output — the
-    reference's headline configuration (38.5 FPS on a 2080 Ti,
-    BASELINE.md / reference test_inference_speed.py). Metric: FPS.
+  * --mode infer (default): forward-only inference, batch 4, last-stack output
+    — the reference's HEADLINE configuration (38.5 FPS on a 2080 Ti,
+    BASELINE.md / reference test_inference_speed.py). Metric: FPS @512x512,
+    vs_baseline = FPS / 38.5.
+  * --mode train: full training step — synthetic batch -> forward -> focal-L2
+    loss -> backward (with overlapped RCCL all-reduce when N > 1) -> fused SGD
+    step. Metric: aggregate train_images_per_sec over all ranks (weak scaling:
+    fixed per-GPU batch). --data device-gt additionally runs the on-device
+    ground-truth generator inside the timed region (the full 512^2 pipeline
+    stays on the GPU).
 
 Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W
@@ -42,7 +45,11 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--mode", choices=["train", "infer"], default="train")
+    p.add_argument("--mode", choices=["train", "infer"], default="infer")
+    p.add_argument("--data", choices=["prestaged", "device-gt"], default="prestaged",
+                   help="train-mode input: pre-staged batches (compute-only "
+                        "timing) or the on-device GT generator inside the "
+                        "timed region")
     p.add_argument("--batch", type=int, default=None,
                    help="per-GPU batch (default: 16 train / 4 infer)")
     p.add_argument("--nstack", type=int, default=4)
@@ -100,16 +107,32 @@ def main():
             from improved_body_parts_amd.parallel import GradReducer
             reducer = GradReducer(model, broadcast_parameters=True)
 
-        # pre-stage a few synthetic batches on device (different per rank)
-        ds = SyntheticPoseDataset(config, length=world_size * 2, seed=17)
-        batches = []
-        for b in range(2):
-            idx = rank * 2 + b
-            img, mm, hm = ds[idx]
-            img = img[None].expand(batch, -1, -1, -1).contiguous()
-            mm = mm[None].expand(batch, -1, -1, -1).contiguous()
-            hm = hm[None].expand(batch, -1, -1, -1).contiguous()
-            batches.append(tuple(t.to(device=device, dtype=dtype) for t in (img, mm, hm)))
+        if args.data == "device-gt" and use_cuda:
+            # the on-device GT pipeline (HIP heatmapper + masks) runs INSIDE
+            # the timed region — what a real training step pays
+            from improved_body_parts_amd.data import DeviceGTSyntheticLoader
+            loader = DeviceGTSyntheticLoader(
+                config, batch, steps_per_epoch=1 << 30, seed=17 + rank,
+                device=device, dtype=dtype)
+            batch_iter = iter(loader)
+
+            def get_batch(i):
+                return next(batch_iter)
+        else:
+            # pre-stage a few synthetic batches on device (different per rank)
+            ds = SyntheticPoseDataset(config, length=world_size * 2, seed=17)
+            batches = []
+            for b in range(2):
+                idx = rank * 2 + b
+                img, mm, hm = ds[idx]
+                img = img[None].expand(batch, -1, -1, -1).contiguous()
+                mm = mm[None].expand(batch, -1, -1, -1).contiguous()
+                hm = hm[None].expand(batch, -1, -1, -1).contiguous()
+                batches.append(tuple(t.to(device=device, dtype=dtype)
+                                     for t in (img, mm, hm)))
+
+            def get_batch(i):
+                return batches[i % len(batches)]
 
         def train_step(b):
             if reducer is not None:
@@ -128,6 +151,8 @@ def main():
             optimizer.step()
             return loss
 
+        if args.graph and args.data == "device-gt":
+            raise SystemExit("--graph requires pre-staged data (--data prestaged)")
         if args.graph and use_cuda and not distributed:
             # whole-step hipGraph: fwd + focal-L2 + bwd + fused SGD captured
             # once, replayed per step (there is no host sync inside the step)
@@ -147,7 +172,7 @@ def main():
                 return static_loss
         else:
             def step(i):
-                return train_step(batches[i % len(batches)])
+                return train_step(get_batch(i))
 
         metric_name = "train_images_per_sec"
         vs_baseline = None
@@ -234,6 +259,7 @@ def main():
             "config": {
                 "model": f"{args.nstack}-stage IMHN @{args.input}x{args.input}",
                 "mode": args.mode,
+                "train_data": args.data if args.mode == "train" else None,
                 "global_batch": batch * world_size,
                 "input": args.input,
                 "parallelism": f"dp{world_size}",

@@ -151,6 +151,10 @@ class DeviceGTSyntheticLoader:
             joints = np.full((self.batch_size, self.max_people,
                               cfg.num_parts, 3), 2.0, dtype=np.float32)
             masks = np.zeros((self.batch_size, h, w), dtype=np.float32)
+            # mask_miss: random unannotated patches, same distribution as
+            # SyntheticPoseDataset — device-GT training exercises the
+            # masked-loss path too
+            mask_miss_np = np.ones((self.batch_size, 1, h, w), dtype=np.float32)
             for b in range(self.batch_size):
                 people = sample_people(rng, cfg.width, cfg.height,
                                        self.max_people)
@@ -164,12 +168,17 @@ class DeviceGTSyntheticLoader:
                     x0, x1 = int(max(xs.min() - 2, 0)), int(min(xs.max() + 2, w))
                     y0, y1 = int(max(ys.min() - 2, 0)), int(min(ys.max() + 2, h))
                     masks[b, y0:y1, x0:x1] = 1.0
+                if rng.random() < 0.5 and w >= 12 and h >= 12:
+                    mw = int(rng.integers(2, max(w // 4, 3)))
+                    mh = int(rng.integers(2, max(h // 4, 3)))
+                    mx = int(rng.integers(0, w - mw))
+                    my = int(rng.integers(0, h - mh))
+                    mask_miss_np[b, 0, my:my + mh, mx:mx + mw] = 0.0
             images = torch.rand(self.batch_size, cfg.height, cfg.width, 3,
                                 generator=gen, device=self.device,
                                 dtype=torch.float32)
             heatmaps = create_heatmaps_device(joints, masks, cfg,
                                               device=self.device)
-            mask_miss = torch.ones(self.batch_size, 1, h, w,
-                                   device=self.device)
+            mask_miss = torch.from_numpy(mask_miss_np).to(self.device)
             yield (images.to(self.dtype), mask_miss.to(self.dtype),
                    heatmaps.to(self.dtype))

@@ -294,6 +294,10 @@ def find_connections(all_peaks, paf_avg, image_height, params, config: Canonical
         for (kk, i, j), (s_prior, pass_ratio, length) in zip(meta, block):
             pa = all_peaks[a_part][i]
             pb = all_peaks[b_part][j]
+            # coincident peaks of two part types form a zero-length segment;
+            # the reference rejects them outright (evaluate.py:229 norm == 0)
+            if length < 1e-6:
+                continue
             # criterion1: enough samples above thre2; criterion2: positive score
             if pass_ratio >= connect_ration and s_prior > 0:
                 combined = 0.5 * s_prior + 0.25 * pa[2] + 0.25 * pb[2]

@@ -119,7 +119,7 @@ class Trainer:
             from ..data import DeviceGTSyntheticLoader
             self.train_sampler = None
             self.train_loader = DeviceGTSyntheticLoader(
-                train_dataset if isinstance(train_dataset, int) else config,
+                config,
                 batch_size=opt.batch_size, steps_per_epoch=device_synth_steps,
                 seed=rank + 1, device=self.device,
                 dtype=torch.bfloat16 if self.bf16 else torch.float32)

@@ -265,7 +265,18 @@ Tensor se_gate(const Tensor& pooled, const Tensor& w1, const Tensor& b1,
   int64_t N = pooled.size(0), C = pooled.size(1);
   int64_t CH = w1.size(0);
   TORCH_CHECK(C <= 768 && CH <= 64, "se_gate LDS layout limit");
-  TORCH_CHECK(w1.is_contiguous() && w2.is_contiguous());
+  TORCH_CHECK(w1.is_contiguous() && w2.is_contiguous() &&
+              b1.is_contiguous() && b2.is_contiguous());
+  // all four params are reinterpreted as w1's dtype below — a mixed-dtype
+  // SE module would silently read garbage without these checks
+  TORCH_CHECK(b1.scalar_type() == w1.scalar_type() &&
+              w2.scalar_type() == w1.scalar_type() &&
+              b2.scalar_type() == w1.scalar_type(),
+              "se_gate: w1/b1/w2/b2 dtypes must match");
+  TORCH_CHECK(w1.dim() == 2 && w1.size(1) == C && b1.numel() == CH &&
+              w2.dim() == 2 && w2.size(0) == C && w2.size(1) == CH &&
+              b2.numel() == C,
+              "se_gate: expected w1 [CH,C], b1 [CH], w2 [C,CH], b2 [C]");
   Tensor s = torch::empty({N, C}, pooled.options());
   dim3 block(256), grid((unsigned)N);
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,

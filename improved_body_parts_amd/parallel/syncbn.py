@@ -26,19 +26,25 @@ class SyncBatchNorm2d(nn.BatchNorm2d):
         if not self.training or world == 1:
             return super().forward(x)
 
+        # fused (sum, sumsq, count) exchange: ONE collective per layer, routed
+        # through the DIFFERENTIABLE all_reduce so backward carries the
+        # cross-rank dmean/dvar terms (its backward all-reduces the stats
+        # gradient). An in-place dist.all_reduce here would be invisible to
+        # autograd and silently drop those terms.
+        import torch.distributed.nn.functional as dist_nn
+        C = self.num_features
         xf = x.float()
         n_local = x.numel() // x.shape[1]
-        # fused (sum, sumsq, count) exchange: one collective per layer
-        stats = torch.empty(2 * self.num_features + 1, dtype=torch.float32,
-                            device=x.device)
-        stats[:self.num_features] = xf.sum(dim=(0, 2, 3))
-        stats[self.num_features:2 * self.num_features] = (xf * xf).sum(dim=(0, 2, 3))
-        stats[-1] = float(n_local)
-        dist.all_reduce(stats, op=dist.ReduceOp.SUM, group=self.process_group)
-        count = stats[-1].clamp(min=1.0)
-        mean = stats[:self.num_features] / count
-        var = stats[self.num_features:2 * self.num_features] / count - mean * mean
-        var = var.clamp(min=0.0)
+        local = torch.cat([
+            xf.sum(dim=(0, 2, 3)),
+            (xf * xf).sum(dim=(0, 2, 3)),
+            torch.full((1,), float(n_local), dtype=torch.float32, device=x.device),
+        ])
+        stats = dist_nn.all_reduce(local, op=dist.ReduceOp.SUM,
+                                   group=self.process_group)
+        count = stats[-1].detach().clamp(min=1.0)  # data-independent scalar
+        mean = stats[:C] / count
+        var = (stats[C:2 * C] / count - mean * mean).clamp(min=0.0)
 
         if self.track_running_stats:
             with torch.no_grad():

@@ -35,6 +35,10 @@ def test_bench_single_process_json_schema():
     assert r.returncode == 0, r.stderr[-2000:]
     j = _last_json_line(r.stdout)
     assert REQUIRED_FIELDS.issubset(j.keys()), j.keys()
+    # the DEFAULT invocation (what the driver runs) emits the inference FPS
+    # headline with vs_baseline filled
+    assert j["metric"] == "fps_512_infer"
+    assert j["vs_baseline"] is not None
     assert j["n_gpus"] == 1 and j["steps"] == 1
     assert j["data"] == "synthetic"
     assert isinstance(j["config"], dict) and "global_batch" in j["config"]
