@@ -120,14 +120,26 @@ def conv_dgrad(dy, weight, x_shape, stride, padding, dilation):
 
 
 def conv_wgrad(x, dy, w_shape, stride, padding, dilation):
-    """Weight gradient stays on the library igemm path (returns None).
+    """Hand-written MFMA weight gradient (csrc/conv_wgrad.hip).
 
-    Deliberate: wgrad's contraction runs over M = N*Ho*Wo, so BOTH MFMA
-    operands need m-contiguous fragments while x/dy are channel-contiguous
-    NHWC — every formulation transposes in LDS. The only fast transposed read
-    on gfx950 is `ds_read_b64_tr_b16` (inline-asm only, exact subtile layout
-    required; no ISA doc in this environment to write it against), and the
-    library's igemm_wrw already measures ~450 TF here — about the same as our
-    own forward kernel on the matching shapes (profiles/). A hand-written
-    wgrad would trade high bring-up risk for ~0 measured headroom."""
-    return None
+    The contraction runs over M = N*Ho*Wo — the strided dimension of both
+    NHWC operands — so fragments are staged through LDS [64 m][16 ch]
+    subtiles with a permuted row order and consumed with gfx950's hardware
+    transpose read ``ds_read_b64_tr_b16`` (round 1 left this on MIOpen
+    igemm_wrw; the builtin ``__builtin_amdgcn_ds_read_tr16_b64_v4i16``
+    makes the layout tractable). Any KHxKW/stride/dilation; bf16 only."""
+    if DISABLE or x.dtype != torch.bfloat16 or dy.dtype != torch.bfloat16:
+        return None
+    if stride[0] != stride[1]:
+        return None
+    ext = hip_extension()
+    if not hasattr(ext, "conv_mfma_wgrad"):
+        return None
+    x = x.contiguous(memory_format=_CL)
+    dy = dy.contiguous(memory_format=_CL)
+    n, cin, h, w_ = x.shape
+    cout, _, kh, kw = w_shape
+    ho, wo = dy.shape[2], dy.shape[3]
+    return ext.conv_mfma_wgrad(x, dy, n, h, w_, cin, cout, kh, kw,
+                               stride[0], padding[0], padding[1],
+                               dilation[0], dilation[1], ho, wo)

@@ -58,6 +58,13 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
                      const c10::optional<Tensor>& shift,
                      const c10::optional<Tensor>& residual, bool act);
 
+// conv_wgrad.hip
+Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
+                       int64_t W, int64_t Cin, int64_t Cout, int64_t KH,
+                       int64_t KW, int64_t stride, int64_t pad_h, int64_t pad_w,
+                       int64_t dil_h, int64_t dil_w, int64_t Ho, int64_t Wo);
+Tensor tr16_probe();
+
 // sgd.hip
 void fused_sgd(const Tensor& chunk_table, double lr, double momentum,
                double weight_decay, int64_t dtype_tag);
@@ -101,6 +108,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dil_h"), py::arg("dil_w"), py::arg("Ho"), py::arg("Wo"),
         py::arg("scale") = c10::nullopt, py::arg("shift") = c10::nullopt,
         py::arg("residual") = c10::nullopt, py::arg("act") = false);
+  m.def("conv_mfma_wgrad", &conv_mfma_wgrad,
+        "MFMA implicit-GEMM conv weight gradient (tr16 LDS transpose)");
+  m.def("tr16_probe", &tr16_probe,
+        "debug: ds_read_b64_tr_b16 delivery-map probe");
   m.def("fused_sgd", &fused_sgd);
   m.def("heatmap_gt", &heatmap_gt, "on-device GT heatmap/PAF generation",
         py::arg("joints"), py::arg("mask_all"), py::arg("limb_pairs"),

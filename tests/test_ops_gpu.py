@@ -180,6 +180,55 @@ def test_mfma_conv_dgrad_vs_reference():
         _assert_rel(dx, ref, 1e-2, f"dgrad {(cin, cout, hw, k, d)}")
 
 
+def test_tr16_probe_delivery_map():
+    """Pin the ds_read_b64_tr_b16 semantics the wgrad kernel is built on:
+    with per-lane address base + l*8B over a [64][16]-short LDS image, lane l
+    elem j receives image element (l&15) + j*16 + (l>>4)*64."""
+    ext = _backend.hip_extension()
+    out = ext.tr16_probe().cpu()  # [64, 4] int16
+    for lane in range(64):
+        for j in range(4):
+            expect = (lane & 15) + j * 16 + (lane >> 4) * 64
+            assert int(out[lane, j]) == expect, \
+                f"lane {lane} elem {j}: got {int(out[lane, j])}, want {expect}"
+
+
+def test_mfma_conv_wgrad_vs_reference():
+    """Hand-written wgrad (tr16-transposed MFMA) vs torch.nn.grad.conv2d_weight
+    across the IMHN's shapes, incl. tap-crossing tails and the 7x7 s2 stem."""
+    from improved_body_parts_amd.ops import conv_kernels
+    ext = _backend.hip_extension()
+    if not hasattr(ext, "conv_mfma_wgrad"):
+        pytest.skip("wgrad kernel not built yet")
+    cases = [
+        # (n, cin, cout, hw, k, s, d)
+        (2, 64, 64, 64, 1, 1, 1),     # 1x1
+        (2, 64, 128, 32, 3, 1, 1),    # 3x3
+        (1, 128, 128, 128, 3, 1, 3),  # dilated 3
+        (1, 128, 128, 64, 3, 1, 5),   # dilated 5
+        (2, 384, 384, 8, 3, 1, 1),    # small spatial, wide channels
+        (2, 256, 50, 32, 1, 1, 1),    # head 1x1 to 50ch (cout tail)
+        (1, 50, 256, 32, 1, 1, 1),    # merge 1x1 from 50ch (cin tail)
+        (2, 3, 64, 64, 7, 2, 1),      # 7x7 s2 stem (elementwise gather)
+        (3, 192, 320, 20, 3, 1, 1),   # odd M tail
+        (2, 16, 64, 32, 4, 1, 1),     # Cin=16 multi-tap (s2d stem shape)
+    ]
+    for n, cin, cout, hw, k, s, d in cases:
+        torch.manual_seed(3)
+        x = torch.randn(n, cin, hw, hw, device="cuda").bfloat16() \
+            .contiguous(memory_format=CL)
+        pad = (k - 1) // 2 * d
+        ho = (hw + 2 * pad - d * (k - 1) - 1) // s + 1
+        dy = (torch.randn(n, cout, ho, ho, device="cuda") * 0.1).bfloat16() \
+            .contiguous(memory_format=CL)
+        dw = conv_kernels.conv_wgrad(x, dy, (cout, cin, k, k), (s, s),
+                                     (pad, pad), (d, d))
+        assert dw is not None, f"wgrad not covered: {(n, cin, cout, hw, k, s, d)}"
+        ref = torch.nn.grad.conv2d_weight(x.float(), (cout, cin, k, k),
+                                          dy.float(), s, pad, d)
+        _assert_rel(dw, ref, 1.5e-2, f"wgrad {(n, cin, cout, hw, k, s, d)}")
+
+
 # ---------------------------------------------------------------------------
 # spatial ops
 # ---------------------------------------------------------------------------
