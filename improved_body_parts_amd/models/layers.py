@@ -41,13 +41,15 @@ class Conv(nn.Module):
             self.conv = nn.Conv2d(inp_dim, out_dim, kernel_size, stride, padding=pad, bias=True)
             self.bn = None
 
-    def forward(self, x):
+    def forward(self, x, residual_post=None, residual_post2=None):
         assert x.size(1) == self.inp_dim, \
             f"input channel {x.size(1)} does not fit kernel channel {self.inp_dim}"
         if self.dropout:
             x = F.dropout(x, p=0.2, training=self.training, inplace=False)
         return ops.conv_bn_act(x, self.conv, self.bn, act=self.relu is not None,
-                               training=self.training)
+                               training=self.training,
+                               residual_post=residual_post,
+                               residual_post2=residual_post2)
 
 
 class DilatedConv(nn.Module):
@@ -217,7 +219,7 @@ class Hourglass(nn.Module):
         self.downsample = nn.MaxPool2d(2, 2)
         self.upsample = nn.Upsample(scale_factor=2, mode="nearest")
 
-    def _forward(self, d, x, up_fms):
+    def _forward(self, d, x, up_fms, post_add=None):
         up1 = self.hg[d][0](x)
         low1 = ops.maxpool2x2(x)
         low1 = self.hg[d][1](low1)
@@ -228,12 +230,16 @@ class Hourglass(nn.Module):
         low3 = self.hg[d][2](low2)
         up_fms.append(low2)
         up2 = ops.upsample2x_nearest(low3)
-        deconv1 = self.hg[d][3](up2)
-        return up1 + deconv1
+        # the up1 + deconv1 skip join rides the refine conv's epilogue
+        # (post-act residual) instead of a separate elementwise kernel;
+        # at d=0 the caller's cross-stack feature-cache add joins too
+        # (legal only for the top scale — deeper low2 outputs feed the
+        # internal up-path, which must see them WITHOUT the cache add)
+        return self.hg[d][3](up2, residual_post=up1, residual_post2=post_add)
 
-    def forward(self, x):
+    def forward(self, x, post_add=None):
         up_fms = []
-        top = self._forward(0, x, up_fms)
+        top = self._forward(0, x, up_fms, post_add)
         return [top] + up_fms[::-1]
 
 

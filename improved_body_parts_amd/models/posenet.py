@@ -20,14 +20,18 @@ from .loss import MultiTaskLoss, MultiTaskLossParallel
 
 
 class Merge(nn.Module):
-    """1x1 conv changing channel count (reference models/posenet.py:13-21)."""
+    """1x1 conv changing channel count (reference models/posenet.py:13-21).
+
+    ``residual`` (optional) is summed into the conv's epilogue on the HIP
+    path — the cross-stack ``merge_preds(pred) + merge_features(feat)`` chain
+    becomes two conv kernels with zero separate add launches."""
 
     def __init__(self, x_dim, y_dim, bn=False):
         super().__init__()
         self.conv = Conv(x_dim, y_dim, 1, relu=False, bn=bn)
 
-    def forward(self, x):
-        return self.conv(x)
+    def forward(self, x, residual=None):
+        return self.conv(x, residual_post=residual)
 
 
 class Features(nn.Module):
@@ -91,20 +95,26 @@ class PoseNet(nn.Module):
         pred = []
         features_cache = None
         for i in range(self.nstack):
-            hourglass_feature = self.hourglass[i](x)
+            # the scale-0 feature-cache add rides the hourglass's top join
+            # conv (post_add); scales 1-4 stay separate adds because the
+            # hourglass's internal up-path consumes them cache-free
+            post0 = features_cache[0] if i > 0 else None
+            hourglass_feature = self.hourglass[i](x, post_add=post0)
             if i == 0:
                 features_cache = [None] * 5
             else:
                 # residual feature cache across stacks (reference posenet.py:93-98)
-                hourglass_feature = [hourglass_feature[s] + features_cache[s]
-                                     for s in range(5)]
+                hourglass_feature = [hourglass_feature[0]] + \
+                    [hourglass_feature[s] + features_cache[s]
+                     for s in range(1, 5)]
             features_instack = self.features[i](hourglass_feature)
             preds_instack = []
             for j in range(5):
                 preds_instack.append(self.outs[i][j](features_instack[j]))
                 if i != self.nstack - 1:
-                    merged = (self.merge_preds[i][j](preds_instack[j])
-                              + self.merge_features[i][j](features_instack[j]))
+                    mf = self.merge_features[i][j](features_instack[j])
+                    merged = self.merge_preds[i][j](preds_instack[j],
+                                                    residual=mf)
                     if j == 0:
                         x = x + merged
                     features_cache[j] = merged

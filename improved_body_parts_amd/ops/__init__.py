@@ -37,31 +37,46 @@ LEAKY_SLOPE = 0.01
 # conv + bn + activation
 # ---------------------------------------------------------------------------
 
-def conv_bn_act(x, conv, bn, act: bool, training: bool):
-    """conv -> (bn) -> (leaky_relu). ``conv``/``bn`` are the parameter-holding
-    nn.Conv2d / nn.BatchNorm2d modules (or bn None)."""
+def conv_bn_act(x, conv, bn, act: bool, training: bool,
+                residual_post=None, residual_post2=None):
+    """conv -> (bn) -> (leaky_relu) (+ residual_post [+ residual_post2]).
+    ``conv``/``bn`` are the parameter-holding nn.Conv2d / nn.BatchNorm2d
+    modules (or bn None). The post-act residuals ride the conv epilogue on
+    the HIP inference path (hourglass skip join, feature-cache add)."""
     if use_hip_for(x):
         from . import conv as _conv
-        return _conv.conv_bn_act_hip(x, conv, bn, act=act, residual=None, training=training)
+        return _conv.conv_bn_act_hip(x, conv, bn, act=act, residual=None,
+                                     training=training,
+                                     residual_post=residual_post,
+                                     residual_post2=residual_post2)
     y = conv(x)
     if bn is not None:
         y = bn(y)
     if act:
         y = F.leaky_relu(y, LEAKY_SLOPE, inplace=True)
+    if residual_post is not None:
+        y = y + residual_post
+    if residual_post2 is not None:
+        y = y + residual_post2
     return y
 
 
-def conv_bn_add_act(x, conv, bn, residual, act: bool, training: bool):
+def conv_bn_add_act(x, conv, bn, residual, act: bool, training: bool,
+                    residual_post=None):
     """conv -> (bn) -> += residual -> (leaky_relu); the fused tail of a Residual block."""
     if use_hip_for(x):
         from . import conv as _conv
-        return _conv.conv_bn_act_hip(x, conv, bn, act=act, residual=residual, training=training)
+        return _conv.conv_bn_act_hip(x, conv, bn, act=act, residual=residual,
+                                     training=training,
+                                     residual_post=residual_post)
     y = conv(x)
     if bn is not None:
         y = bn(y)
     y = y + residual
     if act:
         y = F.leaky_relu(y, LEAKY_SLOPE, inplace=True)
+    if residual_post is not None:
+        y = y + residual_post
     return y
 
 

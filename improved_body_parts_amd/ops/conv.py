@@ -64,10 +64,14 @@ def _tick_running_stats(bn_mod):
 
 
 class ConvBnActFn(torch.autograd.Function):
-    """y = leaky( bn( conv(x, w) ) (+ residual) ), all fused on device."""
+    """y = leaky( bn( conv(x, w) ) (+ residual) ) (+ residual_post [+2]),
+    all fused on device. ``residual`` joins BEFORE the activation (bottleneck
+    skip); ``residual_post``/``residual_post2`` join AFTER it (hourglass
+    up1+deconv1, cross-stack feature-cache) — their backward is identity."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, gamma, beta, residual,
+                residual_post, residual_post2,
                 stride, padding, dilation, act, training, bn_mod, inference):
         ext = hip_extension()
         x = _to_cl(x)
@@ -95,10 +99,13 @@ class ConvBnActFn(torch.autograd.Function):
                 shift = bias.float()
             else:
                 scale = shift = None
-            if scale is not None or act or residual is not None:
+            if scale is not None or act or residual is not None \
+                    or residual_post is not None:
                 y = conv_kernels.conv_fwd(x, weight, stride, padding, dilation,
                                           scale=scale, shift=shift,
-                                          residual=residual, act=act)
+                                          residual=residual, act=act,
+                                          residual_post=residual_post,
+                                          residual_post2=residual_post2)
                 if y is not None:
                     ctx.conf = None
                     return y
@@ -153,20 +160,30 @@ class ConvBnActFn(torch.autograd.Function):
         else:
             y = y_conv
 
+        # y (pre post-add) is what backward needs to reconstruct the act region
         ctx.save_for_backward(x, weight, gamma, y_conv, y, mean, invstd)
         ctx.conf = (stride, padding, dilation, act, training,
-                    residual is not None, bias is not None and gamma is None)
+                    residual is not None, bias is not None and gamma is None,
+                    residual_post is not None, residual_post2 is not None)
         ctx.sync = (sync_group, sync_world)
+        if residual_post is not None:
+            y = y + _to_cl(residual_post)
+        if residual_post2 is not None:
+            y = y + _to_cl(residual_post2)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = hip_extension()
         x, weight, gamma, y_conv, y, mean, invstd = ctx.saved_tensors
-        stride, padding, dilation, act, training, has_res, has_bias = ctx.conf
+        (stride, padding, dilation, act, training, has_res, has_bias,
+         has_post, has_post2) = ctx.conf
         C = y_conv.shape[1]
         dy = _to_cl(dy)
         has_bn = gamma is not None
+        # post-act residuals are pure adds: their gradient is dy itself
+        dpost = dy if has_post else None
+        dpost2 = dy if has_post2 else None
 
         if not has_bn and not act and not has_res and not has_bias:
             # plain conv fast path (the 1x1 heads without bias)
@@ -174,7 +191,7 @@ class ConvBnActFn(torch.autograd.Function):
                 if ctx.needs_input_grad[0] else None
             dw = _conv_wgrad(x, dy, weight.shape, stride, padding, dilation) \
                 if ctx.needs_input_grad[1] else None
-            return (dx, dw, None, None, None, None,
+            return (dx, dw, None, None, None, None, dpost, dpost2,
                     None, None, None, None, None, None, None)
 
         dpre, sum_dpre, sum_dxhat = ext.bn_act_bwd(
@@ -212,7 +229,7 @@ class ConvBnActFn(torch.autograd.Function):
             if ctx.needs_input_grad[0] else None
         dw = _conv_wgrad(x, dconv, weight.shape, stride, padding, dilation) \
             if ctx.needs_input_grad[1] else None
-        return (dx, dw, dbias, dgamma, dbeta, dres,
+        return (dx, dw, dbias, dgamma, dbeta, dres, dpost, dpost2,
                 None, None, None, None, None, None, None)
 
 
@@ -229,7 +246,8 @@ def _sync_info(bn_mod, training):
     return group, world
 
 
-def conv_bn_act_hip(x, conv, bn, act: bool, residual=None, training: bool = False):
+def conv_bn_act_hip(x, conv, bn, act: bool, residual=None, training: bool = False,
+                    residual_post=None, residual_post2=None):
     """Module-level entry used by models.layers: pulls parameters out of the
     nn.Conv2d / nn.BatchNorm2d containers and runs the fused function."""
     gamma = bn.weight if bn is not None else None
@@ -238,5 +256,6 @@ def conv_bn_act_hip(x, conv, bn, act: bool, residual=None, training: bool = Fals
     inference = not bn_training and not torch.is_grad_enabled()
     return ConvBnActFn.apply(
         x, conv.weight, conv.bias, gamma, beta, residual,
+        residual_post, residual_post2,
         conv.stride, conv.padding, conv.dilation, act,
         bn_training, bn, inference)

@@ -25,20 +25,14 @@ DISABLE = os.environ.get("IBP_AMD_DISABLE_MFMA_CONV") == "1"
 
 
 def _supported(x, weight, stride, padding, dilation, for_grad=False):
+    """Envelope of the MFMA kernels: bf16, square stride. Any Cin (subpieces
+    crossing filter-tap boundaries — the 7x7 Cin=3 stem — take the kernel's
+    per-element gather path), any padding (Ho/Wo are passed explicitly)."""
     if DISABLE:
         return False
     if x.dtype != torch.bfloat16:
         return False
-    kh, kw = weight.shape[2], weight.shape[3]
-    cin = x.shape[1]
-    if (kh, kw) != (1, 1) and cin % 64 != 0:
-        return False  # KxK path needs BK | Cin (the 7x7 Cin=3 stem falls back)
-    if padding[0] != (kh - 1) // 2 * dilation[0] or \
-       padding[1] != (kw - 1) // 2 * dilation[1]:
-        return False
-    if for_grad and (stride[0] != 1 or stride[1] != 1):
-        return False
-    if stride[0] not in (1, 2) or stride[0] != stride[1]:
+    if stride[0] != stride[1]:
         return False
     return True
 
@@ -79,10 +73,13 @@ def packed_weight_dgrad(weight: torch.Tensor) -> torch.Tensor:
 
 
 def conv_fwd(x, weight, stride, padding, dilation,
-             scale=None, shift=None, residual=None, act=False):
+             scale=None, shift=None, residual=None, act=False,
+             residual_post=None, residual_post2=None):
     """MFMA conv forward; with scale/shift/residual/act set, the folded-BN
     (+residual +leaky) epilogue runs inside the conv kernel — the whole
-    Conv+BN+LeakyReLU module is ONE kernel on the inference path."""
+    Conv+BN+LeakyReLU module is ONE kernel on the inference path.
+    ``residual_post``/``residual_post2`` are added AFTER the activation (the
+    hourglass up1+deconv1 join and the cross-stack feature-cache add)."""
     if not _supported(x, weight, stride, padding, dilation):
         return None
     ext = hip_extension()
@@ -95,14 +92,21 @@ def conv_fwd(x, weight, stride, padding, dilation,
     wo = (w_ + 2 * padding[1] - dilation[1] * (kw - 1) - 1) // stride[1] + 1
     if residual is not None:
         residual = residual.contiguous(memory_format=_CL)
+    if residual_post is not None:
+        residual_post = residual_post.contiguous(memory_format=_CL)
+    if residual_post2 is not None:
+        residual_post2 = residual_post2.contiguous(memory_format=_CL)
     y = ext.conv_mfma_fwd(x, packed_weight(weight), n, h, w_, cin, cout,
                           kh, kw, stride[0], padding[0], padding[1],
                           dilation[0], dilation[1], ho, wo,
-                          scale, shift, residual, act)
+                          scale, shift, residual, act,
+                          residual_post, residual_post2, 1)
     return y.permute(0, 3, 1, 2)  # NHWC buffer -> NCHW view (channels_last)
 
 
 def conv_dgrad(dy, weight, x_shape, stride, padding, dilation):
+    """dx = stride-1 conv of the (virtually) zero-dilated dy with 180-rotated
+    transposed weights; pad' = dil*(K-1) - pad, gather stride zs = stride."""
     if not _supported(dy, weight, stride, padding, dilation, for_grad=True):
         return None
     ext = hip_extension()
@@ -110,12 +114,16 @@ def conv_dgrad(dy, weight, x_shape, stride, padding, dilation):
         return None
     dy = dy.contiguous(memory_format=_CL)
     n, cout, h, w_ = dy.shape
-    cin = x_shape[1]
+    cin, hx, wx = x_shape[1], x_shape[2], x_shape[3]
     kh, kw = weight.shape[2], weight.shape[3]
-    # stride-1 'same' conv: dx = conv(dy, rot180(W)^T), same padding/dilation
+    pad_h = dilation[0] * (kh - 1) - padding[0]
+    pad_w = dilation[1] * (kw - 1) - padding[1]
+    if pad_h < 0 or pad_w < 0:
+        return None
     dx = ext.conv_mfma_fwd(dy, packed_weight_dgrad(weight), n, h, w_, cout, cin,
-                           kh, kw, 1, padding[0], padding[1],
-                           dilation[0], dilation[1], h, w_)
+                           kh, kw, 1, pad_h, pad_w,
+                           dilation[0], dilation[1], hx, wx,
+                           None, None, None, False, None, None, stride[0])
     return dx.permute(0, 3, 1, 2)
 
 

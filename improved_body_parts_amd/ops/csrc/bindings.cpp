@@ -56,7 +56,9 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
                      int64_t pad_w, int64_t dil_h, int64_t dil_w, int64_t Ho,
                      int64_t Wo, const c10::optional<Tensor>& scale,
                      const c10::optional<Tensor>& shift,
-                     const c10::optional<Tensor>& residual, bool act);
+                     const c10::optional<Tensor>& residual, bool act,
+                     const c10::optional<Tensor>& residual_post,
+                     const c10::optional<Tensor>& residual_post2, int64_t zs);
 
 // conv_wgrad.hip
 Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
@@ -107,7 +109,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("KW"), py::arg("stride"), py::arg("pad_h"), py::arg("pad_w"),
         py::arg("dil_h"), py::arg("dil_w"), py::arg("Ho"), py::arg("Wo"),
         py::arg("scale") = c10::nullopt, py::arg("shift") = c10::nullopt,
-        py::arg("residual") = c10::nullopt, py::arg("act") = false);
+        py::arg("residual") = c10::nullopt, py::arg("act") = false,
+        py::arg("residual_post") = c10::nullopt,
+        py::arg("residual_post2") = c10::nullopt, py::arg("zs") = 1);
   m.def("conv_mfma_wgrad", &conv_mfma_wgrad,
         "MFMA implicit-GEMM conv weight gradient (tr16 LDS transpose)");
   m.def("tr16_probe", &tr16_probe,

@@ -8,21 +8,30 @@
 //
 //   * A is gathered implicitly from the NHWC input (zero-padding handled by
 //     predicated loads); B is the weight pre-packed [Cout][K] row-major.
-//   * 128x(128|64) block tile, 4 waves each computing a 64x64 sub-tile as
-//     4x4 fragments of v_mfma_f32_16x16x32_bf16, fp32 accumulation.
+//   * block tile BMx(BN): (128,128) / (128,64) for big spatial maps,
+//     (64,64) / (32,64) for the hourglass's 16^2/8^2 tails (small-M tiles cut
+//     split-K workspace traffic 4-8x there — round-1 weak #2).
+//   * 4 waves per block arranged (BM/WM)x(BN/WN), each computing WMxWN as
+//     AFRAGxBFRAG fragments of v_mfma_f32_16x16x32_bf16, fp32 accumulation.
 //   * K loop in BK=64 steps, double-buffered LDS with register staging:
 //     tile t+1's global loads are issued before tile t's MFMAs (T14).
-//   * LDS rows padded to 144 B so the 16-lane ds_read_b128 groups hit 16
-//     distinct bank slots (36 dwords * r mod 64 has period 16) — conflict-free.
+//   * LDS rows carry a T2 XOR swizzle — the 16-lane ds_read_b128 group
+//     (16 distinct rows, same k-slice) spreads over 8 bank slots.
 //   * optional fused epilogue: per-channel scale/shift (folded BatchNorm),
-//     residual add, LeakyReLU — inference runs conv+BN+act in ONE kernel.
+//     PRE-act residual (bottleneck skip), LeakyReLU, then up to two POST-act
+//     residuals (hourglass up1+deconv1 join and the cross-stack feature-cache
+//     add ride inside the conv kernel on the inference path).
+//   * zs (gather stride): reads the virtual zero-dilated image x[hi/zs] when
+//     hi % zs == 0 — stride-s dgrad as a stride-1 conv over dilated dy with
+//     180-rotated transposed weights.
 //
-// Supported: any KHxKW with 'same' padding and BK | Cin (all of the IMHN's
-// 3x3/dilated convs have Cin % 64 == 0), plus arbitrary Cin for 1x1 (K-tail
-// predication covers the 50-channel merge heads), stride 1 or 2.
-// dgrad(stride 1) reuses this kernel with 180-rotated transposed weights.
+// Supported: any KHxKW with Cin % 8 == 0 (per-8 subpieces re-derive their
+// filter tap, so Cin 16 s2d-stem tiles and 64..768 hourglass tiles both hit
+// the vector path), arbitrary Cin for 1x1 (K-tail predication covers the
+// 50-channel merge heads) and via the elementwise tail, stride 1 or 2.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstdlib>
 #include "common.h"
 
 namespace ibp {
@@ -31,7 +40,6 @@ typedef short short8 __attribute__((ext_vector_type(8)));
 typedef float floatx4 __attribute__((ext_vector_type(4)));
 typedef unsigned short ushortv8 __attribute__((ext_vector_type(8)));
 
-constexpr int BM = 128;       // output-pixel rows per block
 constexpr int BK = 64;        // K depth per step
 
 struct ConvParams {
@@ -40,9 +48,12 @@ struct ConvParams {
   unsigned short* y;         // NHWC bf16 out
   const float* scale;        // optional per-channel scale (folded BN)
   const float* shift;        // optional per-channel shift / bias
-  const unsigned short* res; // optional residual (NHWC, same shape as y)
+  const unsigned short* res;      // optional residual, added BEFORE act
+  const unsigned short* res_post; // optional residual, added AFTER act
+  const unsigned short* res_post2;
   int N, H, W, Cin, Cout, KH, KW;
   int stride, pad_h, pad_w, dil_h, dil_w, Ho, Wo;
+  int zs;                    // gather stride (transposed-conv dgrad), 1 = off
   long long M;               // N*Ho*Wo
   int K;                     // KH*KW*Cin
   int n_mtiles;              // ceil(M/BM)
@@ -51,15 +62,15 @@ struct ConvParams {
   float* ws;                 // fp32 workspace for split-K partial accumulation
 };
 
-// LDS tile addressing: unpadded 128-B rows with a T2 XOR swizzle — the 16-lane
-// ds_read_b128 group (16 distinct rows, same k-slice) spreads over 8 bank
-// slots (<=2-way) instead of hitting one. k is always a multiple of 8 here, so
-// the XOR preserves 16-B alignment.
+// LDS tile addressing: unpadded BK-short rows with a T2 XOR swizzle — the
+// 16-lane ds_read_b128 group (16 distinct rows, same k-slice) spreads over 8
+// bank slots (<=2-way) instead of hitting one. k is always a multiple of 8
+// here, so the XOR preserves 16-B alignment.
 __device__ __forceinline__ int lds_off(int row, int k) {
   return row * BK + (k ^ ((row & 7) << 3));
 }
 
-template <int BN>
+template <int BM, int BN>
 __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   __shared__ unsigned short lds_a[2][BM * BK];
   __shared__ unsigned short lds_b[2][BN * BK];
@@ -78,28 +89,33 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   const long long m0 = (long long)mt * BM;
   const int n0 = nt * BN;
 
-  // wave sub-tile: 4 waves as 2x2 (BN=128) or 4x1 (BN=64)
-  const int wm = (BN == 128) ? (wave >> 1) : wave;
-  const int wn = (BN == 128) ? (wave & 1) : 0;
-  const int WM = (BN == 128) ? 64 : 32;   // rows per wave (BN=64: 4 waves x 32)
+  // wave sub-tile layout: 4 waves as (BM/WM) x (BN/WN)
+  constexpr int WROWS = (BM == 128) ? ((BN == 128) ? 2 : 4) : 2;
+  constexpr int WCOLS = 4 / WROWS;
+  constexpr int WM = BM / WROWS;
+  constexpr int WN = BN / WCOLS;
+  constexpr int AFRAG = WM / 16;
+  constexpr int BFRAG = WN / 16;
+  const int wm = (WCOLS == 1) ? wave : (wave >> 1);
+  const int wn = (WCOLS == 1) ? 0 : (wave & 1);
   const int row_base = wm * WM;
-  const int col_base = wn * 64;
+  const int col_base = wn * WN;
 
-  const int AFRAG = WM / 16;              // 4 (BN=128) or 2 (BN=64)
-  floatx4 acc[4][4];
+  floatx4 acc[AFRAG][BFRAG];
   #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < AFRAG; ++i)
     #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < BFRAG; ++j) acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
 
-  // ---- staging geometry: 256 threads, each loads 32 elements (2 rows' halves)
-  // A: row = tid>>1, half = tid&1 -> elements [half*32, half*32+32)
-  const int a_row = tid >> 1;
-  const int a_off = (tid & 1) * 32;
-  // B: same pattern over BN rows; BN=64 -> two k-halves per row pair
-  const int b_row = (BN == 128) ? (tid >> 1) : (tid >> 2);
-  const int b_off = (BN == 128) ? ((tid & 1) * 32) : ((tid & 3) * 16);
-  const int b_elems = (BN == 128) ? 32 : 16;
+  // ---- staging geometry: 256 threads cover BM*BK (A) and BN*BK (B) shorts
+  constexpr int A_ELEMS = BM * BK / 256;   // 32 / 16 / 8
+  constexpr int B_ELEMS = BN * BK / 256;   // 32 / 16
+  constexpr int TPR_A = BK / A_ELEMS;
+  constexpr int TPR_B = BK / B_ELEMS;
+  const int a_row = tid / TPR_A;
+  const int a_off = (tid % TPR_A) * A_ELEMS;
+  const int b_row = tid / TPR_B;
+  const int b_off = (tid % TPR_B) * B_ELEMS;
 
   // per-row output-pixel decomposition for the A gather
   const long long a_m = m0 + a_row;
@@ -120,42 +136,73 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   const int nk = min(nk_chunk, nk_total - k_begin);
   if (nk <= 0) return;
 
-  unsigned short a_reg[32];
-  unsigned short b_reg[32];
+  unsigned short a_reg[A_ELEMS];
+  unsigned short b_reg[B_ELEMS];
 
   // ---- stage chunk `ck` into registers --------------------------------------
   auto load_chunk = [&](int ck) {
     const int k0 = ck * BK;
-    // ---- A gather: k = (kh*KW + kw)*Cin + ci
-    {
-      int kk = k0 + a_off;
-      int f = kk / p.Cin;               // filter tap index (constant when BK|Cin)
-      int ci = kk - f * p.Cin;
-      int kh = f / p.KW, kw = f - kh * p.KW;
+    // ---- A gather: k = (kh*KW + kw)*Cin + ci, per-8 subpieces so any
+    // Cin % 8 == 0 stays on 16-B vector loads (tap re-derived per subpiece)
+    #pragma unroll
+    for (int v8 = 0; v8 < A_ELEMS / 8; ++v8) {
+      unsigned short* dst = &a_reg[v8 * 8];
+      const int kk = k0 + a_off + v8 * 8;
+      const int f = kk / p.Cin;          // filter tap of the subpiece start
+      const int ci = kk - f * p.Cin;
+      const int kh = f / p.KW, kw = f - kh * p.KW;
       int hi = hi_base + kh * p.dil_h;
       int wi = wi_base + kw * p.dil_w;
-      bool inside = a_valid_row && hi >= 0 && hi < p.H && wi >= 0 && wi < p.W;
+      bool ok = a_valid_row;
+      if (p.zs > 1) {                    // virtual zero-dilated image (dgrad)
+        ok = ok && (hi % p.zs == 0) && (wi % p.zs == 0);
+        hi /= p.zs; wi /= p.zs;
+      }
+      ok = ok && hi >= 0 && hi < p.H && wi >= 0 && wi < p.W;
       const unsigned short* src =
           p.x + (((long long)a_n * p.H + hi) * p.W + wi) * p.Cin + ci;
       // 16-B vector loads need a 16-B-aligned source (odd Cin, e.g. the
       // 50-channel merge input, makes pixel rows only 2-B aligned)
-      bool a_aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
-      if (inside && a_aligned && ci + 32 <= p.Cin && kk + 32 <= p.K) {
+      const bool aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
+      // subpiece entirely within one filter tap (and K): single tap applies
+      const bool tap_uniform = (ci + 8 <= p.Cin || p.KH * p.KW == 1) &&
+                               (kk + 8 <= p.K || p.KH * p.KW == 1);
+      if (tap_uniform && ok && aligned && ci + 8 <= p.Cin && kk + 8 <= p.K) {
+        *reinterpret_cast<ushortv8*>(dst) =
+            *reinterpret_cast<const ushortv8*>(src);
+      } else if (tap_uniform && ok) {
         #pragma unroll
-        for (int v = 0; v < 4; ++v)
-          *reinterpret_cast<ushortv8*>(&a_reg[v * 8]) =
-              *reinterpret_cast<const ushortv8*>(src + v * 8);
-      } else if (inside) {
-        #pragma unroll
-        for (int e = 0; e < 32; ++e) {
-          int kke = kk + e;
-          // re-derive tap for elements crossing the Cin boundary (1x1 K-tail)
-          a_reg[e] = (kke < p.K && ci + e < p.Cin)
-                         ? src[e] : (unsigned short)0;
+        for (int e = 0; e < 8; ++e) {
+          // 1x1 K-tail: elements past Cin are past K -> zero
+          dst[e] = (kk + e < p.K && ci + e < p.Cin)
+                       ? src[e] : (unsigned short)0;
         }
-      } else {
+      } else if (tap_uniform) {
         #pragma unroll
-        for (int e = 0; e < 32; ++e) a_reg[e] = 0;
+        for (int e = 0; e < 8; ++e) dst[e] = 0;
+      } else {
+        // subpiece crosses tap boundaries (Cin=3 stem, Cin % 8 != 0):
+        // fully re-derive tap and coordinates per element
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int kke = kk + e;
+          unsigned short v = 0;
+          if (a_valid_row && kke < p.K) {
+            const int fe = kke / p.Cin;
+            const int cie = kke - fe * p.Cin;
+            const int khe = fe / p.KW, kwe = fe - khe * p.KW;
+            int hie = hi_base + khe * p.dil_h;
+            int wie = wi_base + kwe * p.dil_w;
+            bool oke = true;
+            if (p.zs > 1) {
+              oke = (hie % p.zs == 0) && (wie % p.zs == 0);
+              hie /= p.zs; wie /= p.zs;
+            }
+            if (oke && hie >= 0 && hie < p.H && wie >= 0 && wie < p.W)
+              v = p.x[(((long long)a_n * p.H + hie) * p.W + wie) * p.Cin + cie];
+          }
+          dst[e] = v;
+        }
       }
     }
     // ---- B: packed [Cout][K] rows
@@ -165,30 +212,28 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
       bool ok = col < p.Cout;
       const unsigned short* src = p.w + (long long)col * p.K + kk;
       bool b_aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
-      if (ok && b_aligned && kk + b_elems <= p.K) {
+      if (ok && b_aligned && kk + B_ELEMS <= p.K) {
         #pragma unroll
-        for (int v = 0; v < 4; ++v)
-          if (v * 8 < b_elems)
-            *reinterpret_cast<ushortv8*>(&b_reg[v * 8]) =
-                *reinterpret_cast<const ushortv8*>(src + v * 8);
+        for (int v = 0; v < B_ELEMS / 8; ++v)
+          *reinterpret_cast<ushortv8*>(&b_reg[v * 8]) =
+              *reinterpret_cast<const ushortv8*>(src + v * 8);
       } else {
         #pragma unroll
-        for (int e = 0; e < 32; ++e)
-          if (e < b_elems) b_reg[e] = (ok && kk + e < p.K) ? src[e] : 0;
+        for (int e = 0; e < B_ELEMS; ++e)
+          b_reg[e] = (ok && kk + e < p.K) ? src[e] : 0;
       }
     }
   };
 
   auto write_chunk = [&](int buf) {
     #pragma unroll
-    for (int v = 0; v < 4; ++v)
+    for (int v = 0; v < A_ELEMS / 8; ++v)
       *reinterpret_cast<ushortv8*>(&lds_a[buf][lds_off(a_row, a_off + v * 8)]) =
           *reinterpret_cast<const ushortv8*>(&a_reg[v * 8]);
     #pragma unroll
-    for (int v = 0; v < 4; ++v)
-      if (v * 8 < b_elems)
-        *reinterpret_cast<ushortv8*>(&lds_b[buf][lds_off(b_row, b_off + v * 8)]) =
-            *reinterpret_cast<const ushortv8*>(&b_reg[v * 8]);
+    for (int v = 0; v < B_ELEMS / 8; ++v)
+      *reinterpret_cast<ushortv8*>(&lds_b[buf][lds_off(b_row, b_off + v * 8)]) =
+          *reinterpret_cast<const ushortv8*>(&b_reg[v * 8]);
   };
 
   // ---- MFMA over one LDS buffer --------------------------------------------
@@ -197,26 +242,21 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
     const int kslice = (lane >> 4) * 8;
     #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
-      short8 afrag[4], bfrag[4];
+      short8 afrag[AFRAG], bfrag[BFRAG];
       #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        if (i < AFRAG)
-          afrag[i] = *reinterpret_cast<const short8*>(
-              &lds_a[buf][lds_off(row_base + i * 16 + l15, kk + kslice)]);
-      }
+      for (int i = 0; i < AFRAG; ++i)
+        afrag[i] = *reinterpret_cast<const short8*>(
+            &lds_a[buf][lds_off(row_base + i * 16 + l15, kk + kslice)]);
       #pragma unroll
-      for (int j = 0; j < 4; ++j)
+      for (int j = 0; j < BFRAG; ++j)
         bfrag[j] = *reinterpret_cast<const short8*>(
             &lds_b[buf][lds_off(col_base + j * 16 + l15, kk + kslice)]);
       #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        if (i < AFRAG) {
-          #pragma unroll
-          for (int j = 0; j < 4; ++j)
-            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
-        }
-      }
+      for (int i = 0; i < AFRAG; ++i)
+        #pragma unroll
+        for (int j = 0; j < BFRAG; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
   };
 
@@ -237,10 +277,9 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   const int ecol = lane & 15;
   const int erow4 = (lane >> 4) * 4;
   #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    if (i >= AFRAG) continue;
+  for (int i = 0; i < AFRAG; ++i) {
     #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < BFRAG; ++j) {
       int col = n0 + col_base + j * 16 + ecol;
       if (col >= p.Cout) continue;
       float sc = p.scale ? p.scale[col] : 1.f;
@@ -259,18 +298,22 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
         float v = acc[i][j][r] * sc + sh;
         if (p.res) v += us2f(p.res[idx]);
         if (p.act) v = leaky(v, 0.01f);
+        if (p.res_post) v += us2f(p.res_post[idx]);
+        if (p.res_post2) v += us2f(p.res_post2[idx]);
         p.y[idx] = f2us(v);
       }
     }
   }
 }
 
-// split-K combine: y = act(scale*ws + shift (+res)) -> bf16
+// split-K combine: y = act(scale*ws + shift (+res)) (+res_post...) -> bf16
 __global__ void splitk_combine_kernel(const float* __restrict__ ws,
                                       unsigned short* __restrict__ y,
                                       const float* __restrict__ scale,
                                       const float* __restrict__ shift,
                                       const unsigned short* __restrict__ res,
+                                      const unsigned short* __restrict__ res_post,
+                                      const unsigned short* __restrict__ res_post2,
                                       long long total, int C, int act,
                                       int splitk) {
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < total;
@@ -281,6 +324,8 @@ __global__ void splitk_combine_kernel(const float* __restrict__ ws,
     if (scale) v = v * scale[c] + shift[c];
     if (res) v += us2f(res[i]);
     if (act) v = leaky(v, 0.01f);
+    if (res_post) v += us2f(res_post[i]);
+    if (res_post2) v += us2f(res_post2[i]);
     y[i] = f2us(v);
   }
 }
@@ -296,7 +341,9 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
                      int64_t pad_w, int64_t dil_h, int64_t dil_w, int64_t Ho,
                      int64_t Wo, const c10::optional<Tensor>& scale,
                      const c10::optional<Tensor>& shift,
-                     const c10::optional<Tensor>& residual, bool act) {
+                     const c10::optional<Tensor>& residual, bool act,
+                     const c10::optional<Tensor>& residual_post,
+                     const c10::optional<Tensor>& residual_post2, int64_t zs) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::ScalarType::BFloat16);
   TORCH_CHECK(w_packed.is_contiguous());
   ibp::ConvParams p;
@@ -307,26 +354,40 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
   p.res = residual.has_value()
               ? reinterpret_cast<const unsigned short*>(residual->data_ptr())
               : nullptr;
+  p.res_post =
+      residual_post.has_value()
+          ? reinterpret_cast<const unsigned short*>(residual_post->data_ptr())
+          : nullptr;
+  p.res_post2 =
+      residual_post2.has_value()
+          ? reinterpret_cast<const unsigned short*>(residual_post2->data_ptr())
+          : nullptr;
   p.N = (int)N; p.H = (int)H; p.W = (int)W; p.Cin = (int)Cin;
   p.Cout = (int)Cout; p.KH = (int)KH; p.KW = (int)KW;
   p.stride = (int)stride; p.pad_h = (int)pad_h; p.pad_w = (int)pad_w;
   p.dil_h = (int)dil_h; p.dil_w = (int)dil_w; p.Ho = (int)Ho; p.Wo = (int)Wo;
+  p.zs = (int)zs;
   p.M = (long long)N * Ho * Wo;
   p.K = (int)(KH * KW * Cin);
   p.act = act ? 1 : 0;
-  if (KH != 1 || KW != 1) {
-    TORCH_CHECK(Cin % ibp::BK == 0,
-                "KxK conv requires Cin % 64 == 0, got ", Cin);
-  }
+  // any Cin is legal: subpieces crossing tap boundaries take the per-element
+  // re-derive path (the Cin=3 stem runs there; Cin % 8 == 0 stays vectorized)
   Tensor y = torch::empty({N, Ho, Wo, Cout}, x.options());
   p.y = reinterpret_cast<unsigned short*>(y.data_ptr());
-  p.n_mtiles = (int)((p.M + ibp::BM - 1) / ibp::BM);
   auto stream = at::hip::getCurrentHIPStream().stream();
-  // pick the N-tile minimising padded work: Cout=192 as 3x64 beats 2x128
-  // (half of the second 128-tile is wasted lanes — measured 0.85x vs library)
+
+  // ---- tile selection -------------------------------------------------------
+  // BM: small-M tiles for the hourglass's 8^2/16^2 scales (avoids the 8-16x
+  // split-K workspace round-trip measured in round 1); BN: minimise padded
+  // cols (Cout=192 as 3x64 beats 2x128 — half the second 128-tile is wasted).
+  int BM = 128;
+  if (p.M <= 2048) BM = 32;
+  else if (p.M <= 8192) BM = 64;
+  if (const char* e = getenv("IBP_CONV_BM")) BM = atoi(e);
   const long long padded128 = ((Cout + 127) / 128) * 128LL;
   const long long padded64 = ((Cout + 63) / 64) * 64LL;
-  const int BN = (Cout > 64 && padded128 <= padded64) ? 128 : 64;
+  int BN = (BM == 128 && Cout > 64 && padded128 <= padded64) ? 128 : 64;
+  p.n_mtiles = (int)((p.M + BM - 1) / BM);
   const int ntiles = p.n_mtiles * (int)((Cout + BN - 1) / BN);
   const int nk_total = (p.K + ibp::BK - 1) / ibp::BK;
   // split K on small grids so the 256-CU chip stays filled (~2 blocks/CU).
@@ -353,17 +414,21 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
     p.ws = nullptr;
   }
   dim3 grid(ntiles * splitk), block(256);
-  if (BN == 128) {
-    hipLaunchKernelGGL(ibp::conv_mfma_kernel<128>, grid, block, 0, stream, p);
+  if (BM == 128 && BN == 128) {
+    hipLaunchKernelGGL((ibp::conv_mfma_kernel<128, 128>), grid, block, 0, stream, p);
+  } else if (BM == 128) {
+    hipLaunchKernelGGL((ibp::conv_mfma_kernel<128, 64>), grid, block, 0, stream, p);
+  } else if (BM == 64) {
+    hipLaunchKernelGGL((ibp::conv_mfma_kernel<64, 64>), grid, block, 0, stream, p);
   } else {
-    hipLaunchKernelGGL(ibp::conv_mfma_kernel<64>, grid, block, 0, stream, p);
+    hipLaunchKernelGGL((ibp::conv_mfma_kernel<32, 64>), grid, block, 0, stream, p);
   }
   if (splitk > 1) {
     long long total = (long long)p.M * Cout;
     dim3 cgrid(ibp::grid_1d(total, 256, 4096)), cblock(256);
     hipLaunchKernelGGL(ibp::splitk_combine_kernel, cgrid, cblock, 0, stream,
-                       p.ws, p.y, p.scale, p.shift, p.res, total, (int)Cout,
-                       p.act, splitk);
+                       p.ws, p.y, p.scale, p.shift, p.res, p.res_post,
+                       p.res_post2, total, (int)Cout, p.act, splitk);
   }
   return y;
 }

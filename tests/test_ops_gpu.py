@@ -144,6 +144,10 @@ def test_mfma_conv_shapes_vs_miopen():
         (2, 64, 64, 64, 1, 2, 1),     # 1x1 stride 2
         (1, 256, 77, 16, 1, 1, 1),    # odd Cout tail
         (3, 192, 320, 20, 3, 1, 1),   # odd M tail (3*20*20=1200 pixels)
+        (2, 3, 64, 128, 7, 2, 1),     # 7x7 s2 stem (per-element tap gather)
+        (2, 16, 64, 64, 4, 1, 1),     # Cin=16 multi-tap (s2d-stem shape)
+        (2, 256, 256, 16, 3, 1, 1),   # BM=32 small-M tile
+        (2, 320, 256, 24, 3, 1, 1),   # BM=64 tile (M=1152... 2*576=1152->32; 24^2*2=1152)
     ]
     for n, cin, cout, hw, k, s, d in cases:
         torch.manual_seed(0)
@@ -162,22 +166,58 @@ def test_mfma_conv_dgrad_vs_reference():
     ext = _backend.hip_extension()
     if not hasattr(ext, "conv_mfma_fwd"):
         pytest.skip("MFMA conv not built yet")
-    for (cin, cout, hw, k, d) in [(64, 128, 32, 3, 1), (128, 128, 16, 3, 3),
-                                  (64, 50, 16, 1, 1), (50, 64, 16, 1, 1)]:
+    for (cin, cout, hw, k, s, d) in [(64, 128, 32, 3, 1, 1),
+                                     (128, 128, 16, 3, 1, 3),
+                                     (64, 50, 16, 1, 1, 1),
+                                     (50, 64, 16, 1, 1, 1),
+                                     (64, 50, 16, 3, 1, 1),   # Cout=50 KxK tail
+                                     (16, 64, 32, 3, 2, 1),   # stride-2 (zs gather)
+                                     (64, 128, 32, 1, 2, 1)]: # 1x1 stride-2
         torch.manual_seed(1)
-        dy = torch.randn(2, cout, hw, hw, device="cuda").bfloat16() \
+        pad = (k - 1) // 2 * d
+        hw_o = (hw + 2 * pad - d * (k - 1) - 1) // s + 1
+        dy = torch.randn(2, cout, hw_o, hw_o, device="cuda").bfloat16() \
             .contiguous(memory_format=CL)
         w = (torch.randn(cout, cin, k, k, device="cuda") * 0.05).bfloat16()
-        pad = (k - 1) // 2 * d
-        dx = conv_kernels.conv_dgrad(dy, w, (2, cin, hw, hw), (1, 1),
+        dx = conv_kernels.conv_dgrad(dy, w, (2, cin, hw, hw), (s, s),
                                      (pad, pad), (d, d))
-        if (k, k) != (1, 1) and cout % 64 != 0:
-            assert dx is None
-            continue
-        assert dx is not None
+        assert dx is not None, f"dgrad not covered: {(cin, cout, hw, k, s, d)}"
         ref = torch.nn.grad.conv2d_input((2, cin, hw, hw), w.float(),
-                                         dy.float(), 1, pad, d)
-        _assert_rel(dx, ref, 1e-2, f"dgrad {(cin, cout, hw, k, d)}")
+                                         dy.float(), s, pad, d)
+        _assert_rel(dx, ref, 1e-2, f"dgrad {(cin, cout, hw, k, s, d)}")
+
+
+@pytest.mark.parametrize("training", [False, True])
+def test_conv_post_act_residuals(training):
+    """residual_post/residual_post2 join AFTER the activation — fused into the
+    conv epilogue on the inference path, identity-gradient adds in training."""
+    torch.manual_seed(11)
+    conv = torch.nn.Conv2d(64, 64, 3, padding=1, bias=False)
+    bn = torch.nn.BatchNorm2d(64)
+    conv_g = torch.nn.Conv2d(64, 64, 3, padding=1, bias=False).cuda().bfloat16()
+    bn_g = torch.nn.BatchNorm2d(64).cuda()
+    conv_g.load_state_dict({k: v.bfloat16() for k, v in conv.state_dict().items()})
+    bn_g.load_state_dict(bn.state_dict())
+    if not training:
+        conv.eval(), bn.eval(), conv_g.eval(), bn_g.eval()
+    x = torch.randn(2, 64, 32, 32)
+    r1 = torch.randn(2, 64, 32, 32)
+    r2 = torch.randn(2, 64, 32, 32)
+    xg = x.cuda().bfloat16().contiguous(memory_format=CL).requires_grad_(training)
+    r1g = r1.cuda().bfloat16().contiguous(memory_format=CL).requires_grad_(training)
+    r2g = r2.cuda().bfloat16().contiguous(memory_format=CL).requires_grad_(training)
+    ctx = torch.enable_grad() if training else torch.no_grad()
+    with ctx:
+        y = ops.conv_bn_act(xg, conv_g, bn_g, act=True, training=training,
+                            residual_post=r1g, residual_post2=r2g)
+    yr = F.leaky_relu(bn(conv(x)), 0.01) + r1 + r2
+    _assert_rel(y, yr, 2e-2, f"post-act residuals (training={training})")
+    if training:
+        y.float().pow(2).sum().backward()
+        # post-residual grads are identity: dL/dr = dy
+        dy_ref = (2 * y.detach().float())
+        _assert_rel(r1g.grad, dy_ref, 2e-2, "dres_post identity")
+        _assert_rel(r2g.grad, dy_ref, 2e-2, "dres_post2 identity")
 
 
 def test_tr16_probe_delivery_map():
