@@ -64,13 +64,12 @@ class MultiTaskLoss(nn.Module):
         return sum(loss_scales) / sum(sw) / batch
 
     def _loss_per_scale(self, pred, target):
-        size = pred.shape[-2:]
-        gt = F.adaptive_avg_pool2d(target[1].float(), output_size=size)
-        mask = F.interpolate(target[0].float(), size=size, mode="bilinear", align_corners=False)
-        mask = (mask >= 0.5).to(pred.dtype) * mask.to(pred.dtype)
-        # reference zeroes <0.5 but keeps the (0.5,1] bilinear values
+        # gt/mask go in at FULL resolution: the fused kernel average-pools GT
+        # windows and bilinearly samples + thresholds mask_miss on the fly
+        # (reference loss_model.py:52-56 materialises a pyramid per scale;
+        # the eager fallback inside ops.focal_l2_loss reproduces it exactly)
         return ops.focal_l2_loss(
-            pred, gt.to(pred.dtype), mask,
+            pred, target[1].to(pred.dtype), target[0].to(pred.dtype),
             heat_start=self.heat_start, bkg_start=self.bkg_start,
             gamma=self.gamma,
             multi_task_weight=self.multi_task_weight,

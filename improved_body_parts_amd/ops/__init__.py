@@ -129,13 +129,27 @@ def focal_l2_loss(pred, gt, mask, *, heat_start, bkg_start, gamma=1,
                   nstack_weight=(1, 1, 1, 1), alpha=0.0, beta=0.0):
     """Focal L2 over (nstack, N, C, H, W) predictions.
 
-    ``gt`` is (N, C, H, W); ``mask`` is (N, 1, H, W) mask_miss already resized to
-    this scale. Per-channel task weights: person-mask channel (C-2) gets
-    multi_task_weight, keypoint channels [heat_start, bkg_start) get
-    keypoint_task_weight (reference loss_model.py:146-156).
+    ``gt`` (N, C, H0, W0) and ``mask`` (N, 1, H0, W0) may be at an integer
+    multiple of the prediction resolution: the HIP kernel average-pools GT
+    windows and bilinearly samples mask_miss ON THE FLY (the supervision
+    pyramid never materialises — reference loss_model.py:52-56 builds it
+    per scale with adaptive_avg_pool2d / interpolate). Mask semantics
+    everywhere: bilinear value kept where >= 0.5, else 0. Per-channel task
+    weights: person-mask channel (C-2) gets multi_task_weight, keypoint
+    channels [heat_start, bkg_start) get keypoint_task_weight
+    (reference loss_model.py:146-156).
     """
     if use_hip_for(pred):
         from . import loss as _loss
+        h0, w0 = gt.shape[-2:]
+        h, w = pred.shape[-2:]
+        if h0 % h or w0 % w or h0 // h != w0 // w:
+            # non-integer pyramid ratio (odd input sizes): pre-pool eagerly,
+            # kernel runs with r = 1
+            gt = F.adaptive_avg_pool2d(gt.float(), (h, w)).to(pred.dtype)
+            m = F.interpolate(mask.float(), size=(h, w), mode="bilinear",
+                              align_corners=False)
+            mask = ((m >= 0.5).float() * m).to(pred.dtype)
         return _loss.focal_l2_loss_hip(pred, gt, mask, heat_start=heat_start,
                                        bkg_start=bkg_start, gamma=gamma,
                                        multi_task_weight=multi_task_weight,
@@ -144,6 +158,13 @@ def focal_l2_loss(pred, gt, mask, *, heat_start, bkg_start, gamma=1,
                                        alpha=alpha, beta=beta)
     nstack = pred.shape[0]
     C = pred.shape[2]
+    size = pred.shape[-2:]
+    if gt.shape[-2:] != size:
+        gt = F.adaptive_avg_pool2d(gt.float(), size).to(pred.dtype)
+    m = mask.float()
+    if m.shape[-2:] != size:
+        m = F.interpolate(m, size=size, mode="bilinear", align_corners=False)
+    mask = ((m >= 0.5).float() * m).to(pred.dtype)
     cw = torch.ones(C, dtype=pred.dtype, device=pred.device)
     cw[heat_start:bkg_start] = keypoint_task_weight
     cw[C - 2] = multi_task_weight

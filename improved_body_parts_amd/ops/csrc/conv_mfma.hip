@@ -139,18 +139,33 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   unsigned short a_reg[A_ELEMS];
   unsigned short b_reg[B_ELEMS];
 
-  // ---- stage chunk `ck` into registers --------------------------------------
+  // ---- incremental tap-walking state for the A gather ----------------------
+  // k = (kh*KW + kw)*Cin + ci decodes with ONE division pair at setup; each
+  // chunk then advances (ci, kh, kw) by small wrap steps. The first version
+  // of the per-8 gather re-divided per subpiece (4 integer divisions per
+  // chunk per thread) and measured 0.63-0.68x of the round-1 kernel — the
+  // emulated 32-bit divide beside MFMAs is exactly the anti-lever the guide
+  // warns about.
+  int t_kk = k_begin * BK + a_off;
+  int t_f = t_kk / p.Cin;
+  int t_ci = t_kk - t_f * p.Cin;
+  int t_kh = t_f / p.KW;
+  int t_kw = t_f - t_kh * p.KW;
+  auto tap_advance = [&](int& ci, int& kh, int& kw, int step) {
+    ci += step;
+    while (ci >= p.Cin) {
+      ci -= p.Cin;
+      if (++kw == p.KW) { kw = 0; ++kh; }
+    }
+  };
+
+  // ---- stage the NEXT sequential chunk into registers ----------------------
+  // (must be called with consecutive chunks — the tap state walks forward)
   auto load_chunk = [&](int ck) {
-    const int k0 = ck * BK;
-    // ---- A gather: k = (kh*KW + kw)*Cin + ci, per-8 subpieces so any
-    // Cin % 8 == 0 stays on 16-B vector loads (tap re-derived per subpiece)
+    int kk = t_kk, ci = t_ci, kh = t_kh, kw = t_kw;
     #pragma unroll
     for (int v8 = 0; v8 < A_ELEMS / 8; ++v8) {
       unsigned short* dst = &a_reg[v8 * 8];
-      const int kk = k0 + a_off + v8 * 8;
-      const int f = kk / p.Cin;          // filter tap of the subpiece start
-      const int ci = kk - f * p.Cin;
-      const int kh = f / p.KW, kw = f - kh * p.KW;
       int hi = hi_base + kh * p.dil_h;
       int wi = wi_base + kw * p.dil_w;
       bool ok = a_valid_row;
@@ -164,33 +179,24 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
       // 16-B vector loads need a 16-B-aligned source (odd Cin, e.g. the
       // 50-channel merge input, makes pixel rows only 2-B aligned)
       const bool aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
-      // subpiece entirely within one filter tap (and K): single tap applies
-      const bool tap_uniform = (ci + 8 <= p.Cin || p.KH * p.KW == 1) &&
-                               (kk + 8 <= p.K || p.KH * p.KW == 1);
-      if (tap_uniform && ok && aligned && ci + 8 <= p.Cin && kk + 8 <= p.K) {
+      const bool in_tap = ci + 8 <= p.Cin;
+      if (ok && aligned && in_tap && kk + 8 <= p.K) {
         *reinterpret_cast<ushortv8*>(dst) =
             *reinterpret_cast<const ushortv8*>(src);
-      } else if (tap_uniform && ok) {
+      } else if (ok && in_tap) {
         #pragma unroll
         for (int e = 0; e < 8; ++e) {
-          // 1x1 K-tail: elements past Cin are past K -> zero
-          dst[e] = (kk + e < p.K && ci + e < p.Cin)
-                       ? src[e] : (unsigned short)0;
+          // K-tail (1x1 heads): elements past K are zero
+          dst[e] = (kk + e < p.K) ? src[e] : (unsigned short)0;
         }
-      } else if (tap_uniform) {
-        #pragma unroll
-        for (int e = 0; e < 8; ++e) dst[e] = 0;
-      } else {
+      } else if (!in_tap && a_valid_row) {
         // subpiece crosses tap boundaries (Cin=3 stem, Cin % 8 != 0):
-        // fully re-derive tap and coordinates per element
+        // walk tap/coordinates element by element (no divisions)
+        int cie = ci, khe = kh, kwe = kw;
         #pragma unroll
         for (int e = 0; e < 8; ++e) {
-          const int kke = kk + e;
           unsigned short v = 0;
-          if (a_valid_row && kke < p.K) {
-            const int fe = kke / p.Cin;
-            const int cie = kke - fe * p.Cin;
-            const int khe = fe / p.KW, kwe = fe - khe * p.KW;
+          if (kk + e < p.K) {
             int hie = hi_base + khe * p.dil_h;
             int wie = wi_base + kwe * p.dil_w;
             bool oke = true;
@@ -202,13 +208,25 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
               v = p.x[(((long long)a_n * p.H + hie) * p.W + wie) * p.Cin + cie];
           }
           dst[e] = v;
+          if (++cie == p.Cin) {
+            cie = 0;
+            if (++kwe == p.KW) { kwe = 0; ++khe; }
+          }
         }
+      } else {
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) dst[e] = 0;
       }
+      kk += 8;
+      tap_advance(ci, kh, kw, 8);
     }
+    // walk the persistent state one full chunk forward
+    t_kk += BK;
+    tap_advance(t_ci, t_kh, t_kw, BK);
     // ---- B: packed [Cout][K] rows
     {
       int col = n0 + b_row;
-      int kk = k0 + b_off;
+      int kk = ck * BK + b_off;
       bool ok = col < p.Cout;
       const unsigned short* src = p.w + (long long)col * p.K + kk;
       bool b_aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;

@@ -21,8 +21,11 @@
 //   * row permutation when staging pixel m (within a 32-row block):
 //     imgrow = 16*((m>>2)&1) + 4*(m>>3) + (m&3), so that the two tr reads of a
 //     fragment (base, base+512B) deliver exactly m-slices g*8+0..3 / g*8+4..7.
-//   * block tile: 64 dW-rows x 64 couts, 4 waves as 2x2 of 32x32,
-//     double-buffered 64-m stages.
+//   * block tile TKDxTCO: (128,128) for the big layers (32 MFMA : 32 tr reads
+//     per wave-stage — MFMA-bound) and (64,64) for heads/merges/small KD
+//     (8 MFMA : 16 tr reads), 4 waves as 2x2; double-buffered 64-m stages.
+//   * the per-stage pixel decomposition m -> (n, ho, wo) advances by +64 with
+//     carry steps (no 64-bit divisions in the loop).
 //   * split-M determinism: each m-chunk writes its own fp32 workspace slice
 //     with plain stores; a combine kernel reduces slices in fixed order and
 //     scatters into the [Cout][Cin][KH][KW] weight-grad layout.
@@ -41,10 +44,9 @@ typedef short short8 __attribute__((ext_vector_type(8)));
 typedef float floatx4 __attribute__((ext_vector_type(4)));
 typedef unsigned short ushortv8 __attribute__((ext_vector_type(8)));
 
-// LDS geometry: 8 subtiles (4 A-cin + 4 B-cout), each [64 rows][16 ch] plus
-// a 16-B pad so consecutive subtiles start on different write-bank groups.
+// LDS geometry: (TKD+TCO)/16 subtiles, each [64 rows][16 ch] plus a 16-B pad
+// so consecutive subtiles start on different write-bank groups.
 constexpr int SUB_SHORTS = 64 * 16 + 8;   // 1032 shorts = 2064 B (8-B aligned)
-constexpr int BUF_SHORTS = 8 * SUB_SHORTS;
 
 struct WgradParams {
   const unsigned short* x;    // NHWC bf16
@@ -56,6 +58,7 @@ struct WgradParams {
   int KD;                     // KH*KW*Cin (dW rows)
   int kd_tiles, co_tiles, chunks;
   long long chunk_len;        // multiple of 64
+  int step_dho, step_dwo;     // 64 / Wo, 64 % Wo (m-walk carry steps)
 };
 
 __device__ __forceinline__ int wg_imgrow(int m) {
@@ -69,8 +72,13 @@ __device__ __forceinline__ short4v tr16_read(const unsigned short* p) {
       (__attribute__((address_space(3))) short4v*)p);
 }
 
-__global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(WgradParams p) {
-  __shared__ unsigned short lds[2][BUF_SHORTS];
+template <int TKD, int TCO>
+__global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(WgradParams p) {
+  constexpr int ASUB = TKD / 16;
+  constexpr int BSUB = TCO / 16;
+  constexpr int NSUB = ASUB + BSUB;
+  constexpr int NPIECE = NSUB / 4;        // 16-ch pieces per thread per stage
+  __shared__ unsigned short lds[2][NSUB * SUB_SHORTS];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -81,117 +89,150 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(WgradParams p) {
   const int cot = bid % p.co_tiles; bid /= p.co_tiles;
   const int kdt = bid % p.kd_tiles;
   const int chunk = bid / p.kd_tiles;
-  const int kd0 = kdt * 64;
-  const int co0 = cot * 64;
+  const int kd0 = kdt * TKD;
+  const int co0 = cot * TCO;
   const long long m_begin = (long long)chunk * p.chunk_len;
   const long long m_end = min(m_begin + p.chunk_len, p.M);
   if (m_begin >= m_end) return;
 
-  // wave sub-tile: 2x2 waves of 32x32
+  // wave sub-tile: 2x2 waves of (TKD/2)x(TCO/2)
+  constexpr int AFRAG = TKD / 32;
+  constexpr int BFRAG = TCO / 32;
   const int wm = wave >> 1;       // dW-row half
   const int wn = wave & 1;        // cout half
 
-  floatx4 acc[2][2];
+  floatx4 acc[AFRAG][BFRAG];
   #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < AFRAG; ++i)
     #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < BFRAG; ++j) acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
 
-  // ---- staging geometry: thread t handles pixel m0+(t>>2), subtile q=t&3
+  // ---- staging geometry: thread t handles pixel m0+(t>>2) and NPIECE
+  // subtiles q, q+4, ... (q = t&3); first ASUB subtiles are A, rest B
   const int st_m = tid >> 2;
   const int st_q = tid & 3;
   const int st_row = wg_imgrow(st_m);
 
-  // A-piece dW rows [kd0+16q, +16): tap + cin base (may cross taps only when
-  // the elementwise path is active)
-  const int a_k0 = kd0 + 16 * st_q;
-  const int a_f = p.Cin > 0 ? a_k0 / p.Cin : 0;
-  const int a_ci = a_k0 - a_f * p.Cin;
-  const int a_kh = p.KW > 0 ? a_f / p.KW : 0;
-  const int a_kw = a_f - a_kh * p.KW;
-  // piece crosses a tap boundary (or runs past KD) -> per-element re-derive
-  const bool a_elementwise = (a_ci + 16 > p.Cin) || (a_k0 + 16 > p.KD);
-
-  unsigned short a_reg[16];
-  unsigned short b_reg[16];
-
-  auto load_stage = [&](long long m0) {
-    const long long m = m0 + st_m;
-    int n = 0, ho = 0, wo = 0;
-    bool m_ok = m < p.M;
-    if (m_ok) {
-      long long t = m;
-      wo = (int)(t % p.Wo); t /= p.Wo;
-      ho = (int)(t % p.Ho);
-      n = (int)(t / p.Ho);
-    }
-    // ---- A: x[n, ho*s - pad + kh*dil, wo*s - pad + kw*dil, ci0..ci0+16)
-    if (!a_elementwise) {
-      const int hi = ho * p.stride - p.pad_h + a_kh * p.dil_h;
-      const int wi = wo * p.stride - p.pad_w + a_kw * p.dil_w;
-      const bool inside = m_ok && hi >= 0 && hi < p.H && wi >= 0 && wi < p.W;
-      const unsigned short* src =
-          p.x + (((long long)n * p.H + hi) * p.W + wi) * p.Cin + a_ci;
-      const bool aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
-      if (inside && aligned) {
-        #pragma unroll
-        for (int v = 0; v < 2; ++v)
-          *reinterpret_cast<ushortv8*>(&a_reg[v * 8]) =
-              *reinterpret_cast<const ushortv8*>(src + v * 8);
-      } else if (inside) {
-        #pragma unroll
-        for (int e = 0; e < 16; ++e) a_reg[e] = src[e];
-      } else {
-        #pragma unroll
-        for (int e = 0; e < 16; ++e) a_reg[e] = 0;
-      }
+  // per-piece dW-row bases (tap + cin) — one division pair per piece per block
+  int pc_f[NPIECE], pc_ci[NPIECE];
+  bool pc_elem[NPIECE];     // piece crosses a tap boundary -> per-element path
+  bool pc_isA[NPIECE];
+  int pc_co[NPIECE];
+  #pragma unroll
+  for (int pi = 0; pi < NPIECE; ++pi) {
+    const int sub = st_q + 4 * pi;
+    if (sub < ASUB) {
+      const int k0 = kd0 + 16 * sub;
+      const int f = k0 / p.Cin;
+      pc_isA[pi] = true;
+      pc_f[pi] = f;
+      pc_ci[pi] = k0 - f * p.Cin;
+      pc_elem[pi] = (pc_ci[pi] + 16 > p.Cin) || (k0 + 16 > p.KD);
+      pc_co[pi] = 0;
     } else {
-      #pragma unroll
-      for (int e = 0; e < 16; ++e) {
-        const int k = a_k0 + e;
-        unsigned short v = 0;
-        if (m_ok && k < p.KD) {
-          const int f = k / p.Cin;
-          const int ci = k - f * p.Cin;
-          const int kh = f / p.KW, kw = f - kh * p.KW;
-          const int hi = ho * p.stride - p.pad_h + kh * p.dil_h;
-          const int wi = wo * p.stride - p.pad_w + kw * p.dil_w;
-          if (hi >= 0 && hi < p.H && wi >= 0 && wi < p.W)
-            v = p.x[(((long long)n * p.H + hi) * p.W + wi) * p.Cin + ci];
+      pc_isA[pi] = false;
+      pc_co[pi] = co0 + 16 * (sub - ASUB);
+      pc_f[pi] = pc_ci[pi] = 0;
+      pc_elem[pi] = false;
+    }
+  }
+
+  unsigned short regs[NPIECE][16];
+
+  // ---- incremental pixel decomposition m -> (n, ho, wo): one 64-bit division
+  // pair at setup, then +64 carry steps per stage
+  long long mm = m_begin + st_m;
+  int st_n = 0, st_ho = 0, st_wo = 0;
+  {
+    long long t = mm < p.M ? mm : 0;
+    st_wo = (int)(t % p.Wo); t /= p.Wo;
+    st_ho = (int)(t % p.Ho);
+    st_n = (int)(t / p.Ho);
+  }
+
+  auto load_stage = [&]() {
+    const bool m_ok = mm < p.M;
+    #pragma unroll
+    for (int pi = 0; pi < NPIECE; ++pi) {
+      unsigned short* dst = regs[pi];
+      if (pc_isA[pi]) {
+        if (!pc_elem[pi]) {
+          const int f = pc_f[pi];
+          const int kh = f / p.KW, kw = f - kh * p.KW;   // f small; cheap
+          const int hi = st_ho * p.stride - p.pad_h + kh * p.dil_h;
+          const int wi = st_wo * p.stride - p.pad_w + kw * p.dil_w;
+          const bool inside =
+              m_ok && hi >= 0 && hi < p.H && wi >= 0 && wi < p.W;
+          const unsigned short* src =
+              p.x + (((long long)st_n * p.H + hi) * p.W + wi) * p.Cin +
+              pc_ci[pi];
+          const bool aligned =
+              ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
+          if (inside && aligned) {
+            #pragma unroll
+            for (int v = 0; v < 2; ++v)
+              *reinterpret_cast<ushortv8*>(&dst[v * 8]) =
+                  *reinterpret_cast<const ushortv8*>(src + v * 8);
+          } else if (inside) {
+            #pragma unroll
+            for (int e = 0; e < 16; ++e) dst[e] = src[e];
+          } else {
+            #pragma unroll
+            for (int e = 0; e < 16; ++e) dst[e] = 0;
+          }
+        } else {
+          const int k0 = kd0 + 16 * (st_q + 4 * pi);
+          #pragma unroll
+          for (int e = 0; e < 16; ++e) {
+            const int k = k0 + e;
+            unsigned short v = 0;
+            if (m_ok && k < p.KD) {
+              const int f = k / p.Cin;
+              const int ci = k - f * p.Cin;
+              const int kh = f / p.KW, kw = f - kh * p.KW;
+              const int hi = st_ho * p.stride - p.pad_h + kh * p.dil_h;
+              const int wi = st_wo * p.stride - p.pad_w + kw * p.dil_w;
+              if (hi >= 0 && hi < p.H && wi >= 0 && wi < p.W)
+                v = p.x[(((long long)st_n * p.H + hi) * p.W + wi) * p.Cin + ci];
+            }
+            dst[e] = v;
+          }
         }
-        a_reg[e] = v;
-      }
-    }
-    // ---- B: dy[m, co0 + 16q .. +16)
-    {
-      const int co = co0 + 16 * st_q;
-      const unsigned short* src = p.dy + m * p.Cout + co;
-      const bool full = m_ok && co + 16 <= p.Cout;
-      const bool aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
-      if (full && aligned) {
-        #pragma unroll
-        for (int v = 0; v < 2; ++v)
-          *reinterpret_cast<ushortv8*>(&b_reg[v * 8]) =
-              *reinterpret_cast<const ushortv8*>(src + v * 8);
       } else {
-        #pragma unroll
-        for (int e = 0; e < 16; ++e)
-          b_reg[e] = (m_ok && co + e < p.Cout) ? src[e] : 0;
+        const int co = pc_co[pi];
+        const unsigned short* src = p.dy + mm * p.Cout + co;
+        const bool full = m_ok && co + 16 <= p.Cout;
+        const bool aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
+        if (full && aligned) {
+          #pragma unroll
+          for (int v = 0; v < 2; ++v)
+            *reinterpret_cast<ushortv8*>(&dst[v * 8]) =
+                *reinterpret_cast<const ushortv8*>(src + v * 8);
+        } else {
+          #pragma unroll
+          for (int e = 0; e < 16; ++e)
+            dst[e] = (m_ok && co + e < p.Cout) ? src[e] : 0;
+        }
       }
     }
+    // advance the pixel walker by one stage (+64 pixels)
+    mm += 64;
+    st_wo += p.step_dwo;
+    st_ho += p.step_dho + (st_wo >= p.Wo ? 1 : 0);
+    if (st_wo >= p.Wo) st_wo -= p.Wo;
+    while (st_ho >= p.Ho) { st_ho -= p.Ho; ++st_n; }
   };
 
   auto write_stage = [&](int buf) {
-    unsigned short* a_dst = &lds[buf][st_q * SUB_SHORTS + st_row * 16];
-    unsigned short* b_dst = &lds[buf][(4 + st_q) * SUB_SHORTS + st_row * 16];
     #pragma unroll
-    for (int v = 0; v < 2; ++v)
-      *reinterpret_cast<ushortv8*>(a_dst + v * 8) =
-          *reinterpret_cast<const ushortv8*>(&a_reg[v * 8]);
-    #pragma unroll
-    for (int v = 0; v < 2; ++v)
-      *reinterpret_cast<ushortv8*>(b_dst + v * 8) =
-          *reinterpret_cast<const ushortv8*>(&b_reg[v * 8]);
+    for (int pi = 0; pi < NPIECE; ++pi) {
+      const int sub = st_q + 4 * pi;
+      unsigned short* dst = &lds[buf][sub * SUB_SHORTS + st_row * 16];
+      #pragma unroll
+      for (int v = 0; v < 2; ++v)
+        *reinterpret_cast<ushortv8*>(dst + v * 8) =
+            *reinterpret_cast<const ushortv8*>(&regs[pi][v * 8]);
+    }
   };
 
   // fragment read: subtile s, contraction slice base ks (0/32), via 2 tr reads
@@ -208,15 +249,17 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(WgradParams p) {
   auto compute = [&](int buf) {
     #pragma unroll
     for (int ks = 0; ks < 64; ks += 32) {
-      short8 af[2], bf[2];
+      short8 af[AFRAG], bf[BFRAG];
       #pragma unroll
-      for (int i = 0; i < 2; ++i) af[i] = frag(buf, wm * 2 + i, ks);
+      for (int i = 0; i < AFRAG; ++i)
+        af[i] = frag(buf, wm * AFRAG + i, ks);
       #pragma unroll
-      for (int j = 0; j < 2; ++j) bf[j] = frag(buf, 4 + wn * 2 + j, ks);
+      for (int j = 0; j < BFRAG; ++j)
+        bf[j] = frag(buf, ASUB + wn * BFRAG + j, ks);
       #pragma unroll
-      for (int i = 0; i < 2; ++i)
+      for (int i = 0; i < AFRAG; ++i)
         #pragma unroll
-        for (int j = 0; j < 2; ++j)
+        for (int j = 0; j < BFRAG; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bf[j], acc[i][j], 0, 0, 0);
     }
@@ -224,11 +267,11 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(WgradParams p) {
 
   // ---- main loop over the chunk's m range, register-staged double buffer
   const long long n_stages = (m_end - m_begin + 63) >> 6;
-  load_stage(m_begin);
+  load_stage();
   write_stage(0);
   __syncthreads();
   for (long long t = 0; t < n_stages; ++t) {
-    if (t + 1 < n_stages) load_stage(m_begin + (t + 1) * 64);
+    if (t + 1 < n_stages) load_stage();
     compute((int)(t & 1));
     if (t + 1 < n_stages) write_stage((int)((t + 1) & 1));
     __syncthreads();
@@ -239,14 +282,14 @@ __global__ __launch_bounds__(256, 4) void conv_wgrad_kernel(WgradParams p) {
   const int erow4 = (lane >> 4) * 4;
   float* out = p.ws + (long long)chunk * p.KD * p.Cout;
   #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int i = 0; i < AFRAG; ++i) {
     #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      const int co = co0 + wn * 32 + j * 16 + ecol;
+    for (int j = 0; j < BFRAG; ++j) {
+      const int co = co0 + wn * (TCO / 2) + j * 16 + ecol;
       if (co >= p.Cout) continue;
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int kd = kd0 + wm * 32 + i * 16 + erow4 + r;
+        const int kd = kd0 + wm * (TKD / 2) + i * 16 + erow4 + r;
         if (kd >= p.KD) continue;
         out[(long long)kd * p.Cout + co] = acc[i][j][r];
       }
@@ -307,8 +350,14 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   p.dil_h = (int)dil_h; p.dil_w = (int)dil_w; p.Ho = (int)Ho; p.Wo = (int)Wo;
   p.M = (long long)N * Ho * Wo;
   p.KD = (int)(KH * KW * Cin);
-  p.kd_tiles = (p.KD + 63) / 64;
-  p.co_tiles = (int)((Cout + 63) / 64);
+  p.step_dho = 64 / p.Wo;
+  p.step_dwo = 64 % p.Wo;
+  // big-tile variant when both dims fill it (32 MFMA : 32 tr-reads per
+  // wave-stage); 64x64 otherwise (heads, merges, small KD)
+  const bool big = p.KD >= 128 && Cout >= 128;
+  const int TKD = big ? 128 : 64, TCO = big ? 128 : 64;
+  p.kd_tiles = (p.KD + TKD - 1) / TKD;
+  p.co_tiles = (int)((Cout + TCO - 1) / TCO);
   // split M into chunks for parallelism: aim ~768 blocks, chunk length a
   // multiple of 64; every chunk slice is fully written (plain stores) before
   // the combine reduces them in fixed order -> deterministic
@@ -323,10 +372,16 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   Tensor ws = torch::empty({(long long)p.chunks * p.KD * Cout},
                            x.options().dtype(torch::kFloat32));
   p.ws = ws.data_ptr<float>();
-  // zero only the tiles' dead rows? every in-range (kd, cout) is written by
-  // exactly one block per chunk; out-of-range rows are never read back.
+  // every in-range (kd, cout) is written by exactly one block per chunk;
+  // out-of-range rows are never read back -> no zero-init needed
   dim3 grid(p.chunks * tiles), block(256);
-  hipLaunchKernelGGL(ibp::conv_wgrad_kernel, grid, block, 0, stream, p);
+  if (big) {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128>), grid, block, 0,
+                       stream, p);
+  } else {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<64, 64>), grid, block, 0,
+                       stream, p);
+  }
 
   Tensor dw = torch::empty({Cout, Cin, KH, KW}, x.options());
   long long total = Cout * Cin * KH * KW;
