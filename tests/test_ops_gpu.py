@@ -344,6 +344,30 @@ def test_focal_l2_gpu_vs_cpu(gamma):
     _assert_close(pg.grad, pr.grad, 1e-4, 1e-5, "dpred")
 
 
+@pytest.mark.parametrize("r", [2, 4, 16])
+def test_focal_l2_fused_pyramid(r):
+    """Full-res GT/mask go straight into the kernel, which average-pools GT
+    windows and bilinearly samples + thresholds mask_miss on the fly; vs the
+    CPU eager path (which materialises the pyramid like the reference)."""
+    torch.manual_seed(2)
+    H = 64
+    h = H // r
+    pred = torch.rand(2, 2, 50, h, h)
+    gt = torch.rand(2, 50, H, H) * (torch.rand(2, 50, H, H) > 0.6)
+    mask = (torch.rand(2, 1, H, H) > 0.2).float()
+    kw = dict(heat_start=30, bkg_start=48, gamma=1,
+              multi_task_weight=0.1, keypoint_task_weight=3.0,
+              nstack_weight=(1, 2))
+    pg = pred.cuda().requires_grad_(True)
+    pr = pred.clone().requires_grad_(True)
+    lg = ops.focal_l2_loss(pg, gt.cuda(), mask.cuda(), **kw)
+    lr = ops.focal_l2_loss(pr, gt, mask, **kw)
+    _assert_close(lg, lr, 1e-4, 1e-3 * max(float(lr.detach()), 1.0), "pyr loss")
+    lg.backward()
+    lr.backward()
+    _assert_close(pg.grad, pr.grad, 1e-4, 1e-5, "pyr dpred")
+
+
 def test_focal_l2_bf16():
     torch.manual_seed(1)
     pred = torch.rand(1, 2, 50, 32, 32)
