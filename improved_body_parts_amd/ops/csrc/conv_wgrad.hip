@@ -73,7 +73,8 @@ __device__ __forceinline__ short4v tr16_read(const unsigned short* p) {
 }
 
 template <int TKD, int TCO>
-__global__ __launch_bounds__(256, 2) void conv_wgrad_kernel(WgradParams p) {
+__global__ __launch_bounds__(256, TKD == 64 ? 4 : 2) void conv_wgrad_kernel(
+    WgradParams p) {
   constexpr int ASUB = TKD / 16;
   constexpr int BSUB = TCO / 16;
   constexpr int NSUB = ASUB + BSUB;
@@ -352,9 +353,11 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   p.KD = (int)(KH * KW * Cin);
   p.step_dho = 64 / p.Wo;
   p.step_dwo = 64 % p.Wo;
-  // big-tile variant when both dims fill it (32 MFMA : 32 tr-reads per
-  // wave-stage); 64x64 otherwise (heads, merges, small KD)
-  const bool big = p.KD >= 128 && Cout >= 128;
+  // 64x64 tiles at 4 blocks/CU measured faster than 128x128 at 2 blocks/CU
+  // on every shape (the kernel is load-latency-, not LDS-, bound); the big
+  // variant is kept behind an env knob for tile experiments
+  const char* bt = getenv("IBP_WGRAD_BIG");
+  const bool big = bt && bt[0] == '1' && p.KD >= 128 && Cout >= 128;
   const int TKD = big ? 128 : 64, TCO = big ? 128 : 64;
   p.kd_tiles = (p.KD + TKD - 1) / TKD;
   p.co_tiles = (int)((Cout + TCO - 1) / TCO);
