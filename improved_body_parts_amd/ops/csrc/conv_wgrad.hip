@@ -73,7 +73,7 @@ __device__ __forceinline__ short4v tr16_read(const unsigned short* p) {
 }
 
 template <int TKD, int TCO>
-__global__ __launch_bounds__(256, TKD == 64 ? 4 : 2) void conv_wgrad_kernel(
+__global__ __launch_bounds__(256, TKD == 64 ? 3 : 2) void conv_wgrad_kernel(
     WgradParams p) {
   constexpr int ASUB = TKD / 16;
   constexpr int BSUB = TCO / 16;
@@ -114,8 +114,10 @@ __global__ __launch_bounds__(256, TKD == 64 ? 4 : 2) void conv_wgrad_kernel(
   const int st_q = tid & 3;
   const int st_row = wg_imgrow(st_m);
 
-  // per-piece dW-row bases (tap + cin) — one division pair per piece per block
-  int pc_f[NPIECE], pc_ci[NPIECE];
+  // per-piece dW-row bases (tap + cin) — divisions happen ONCE here; the
+  // steady-state stage loop is division-free (emulated integer divide beside
+  // MFMAs is the anti-lever the guide warns about)
+  int pc_ci[NPIECE], pc_kh[NPIECE], pc_kw[NPIECE];
   bool pc_elem[NPIECE];     // piece crosses a tap boundary -> per-element path
   bool pc_isA[NPIECE];
   int pc_co[NPIECE];
@@ -126,19 +128,25 @@ __global__ __launch_bounds__(256, TKD == 64 ? 4 : 2) void conv_wgrad_kernel(
       const int k0 = kd0 + 16 * sub;
       const int f = k0 / p.Cin;
       pc_isA[pi] = true;
-      pc_f[pi] = f;
       pc_ci[pi] = k0 - f * p.Cin;
+      pc_kh[pi] = f / p.KW;
+      pc_kw[pi] = f - pc_kh[pi] * p.KW;
       pc_elem[pi] = (pc_ci[pi] + 16 > p.Cin) || (k0 + 16 > p.KD);
       pc_co[pi] = 0;
     } else {
       pc_isA[pi] = false;
       pc_co[pi] = co0 + 16 * (sub - ASUB);
-      pc_f[pi] = pc_ci[pi] = 0;
+      pc_ci[pi] = pc_kh[pi] = pc_kw[pi] = 0;
       pc_elem[pi] = false;
     }
   }
 
-  unsigned short regs[NPIECE][16];
+  // two register banks -> loads are issued TWO stages ahead of their LDS
+  // write (PMC r2: 58.7% wait at one-stage prefetch — global latency was not
+  // covered by a single compute phase). Separate named arrays + reference
+  // parameters keep them in registers (a regs[bank] runtime index would
+  // spill the array to scratch).
+  unsigned short regs0[NPIECE][16], regs1[NPIECE][16];
 
   // ---- incremental pixel decomposition m -> (n, ho, wo): one 64-bit division
   // pair at setup, then +64 carry steps per stage
@@ -151,17 +159,15 @@ __global__ __launch_bounds__(256, TKD == 64 ? 4 : 2) void conv_wgrad_kernel(
     st_n = (int)(t / p.Ho);
   }
 
-  auto load_stage = [&]() {
+  auto load_stage = [&](unsigned short (&regs)[NPIECE][16]) {
     const bool m_ok = mm < p.M;
     #pragma unroll
     for (int pi = 0; pi < NPIECE; ++pi) {
       unsigned short* dst = regs[pi];
       if (pc_isA[pi]) {
         if (!pc_elem[pi]) {
-          const int f = pc_f[pi];
-          const int kh = f / p.KW, kw = f - kh * p.KW;   // f small; cheap
-          const int hi = st_ho * p.stride - p.pad_h + kh * p.dil_h;
-          const int wi = st_wo * p.stride - p.pad_w + kw * p.dil_w;
+          const int hi = st_ho * p.stride - p.pad_h + pc_kh[pi] * p.dil_h;
+          const int wi = st_wo * p.stride - p.pad_w + pc_kw[pi] * p.dil_w;
           const bool inside =
               m_ok && hi >= 0 && hi < p.H && wi >= 0 && wi < p.W;
           const unsigned short* src =
@@ -182,21 +188,25 @@ __global__ __launch_bounds__(256, TKD == 64 ? 4 : 2) void conv_wgrad_kernel(
             for (int e = 0; e < 16; ++e) dst[e] = 0;
           }
         } else {
+          // tap-crossing piece (stem, Cin % 16 != 0): walk (ci, kh, kw)
+          // element by element — division-free, NOT unrolled (an unrolled
+          // 16-wide division chain ballooned VGPR pressure into spills)
           const int k0 = kd0 + 16 * (st_q + 4 * pi);
-          #pragma unroll
+          int cie = pc_ci[pi], khe = pc_kh[pi], kwe = pc_kw[pi];
+          #pragma unroll 1
           for (int e = 0; e < 16; ++e) {
-            const int k = k0 + e;
             unsigned short v = 0;
-            if (m_ok && k < p.KD) {
-              const int f = k / p.Cin;
-              const int ci = k - f * p.Cin;
-              const int kh = f / p.KW, kw = f - kh * p.KW;
-              const int hi = st_ho * p.stride - p.pad_h + kh * p.dil_h;
-              const int wi = st_wo * p.stride - p.pad_w + kw * p.dil_w;
+            if (m_ok && k0 + e < p.KD) {
+              const int hi = st_ho * p.stride - p.pad_h + khe * p.dil_h;
+              const int wi = st_wo * p.stride - p.pad_w + kwe * p.dil_w;
               if (hi >= 0 && hi < p.H && wi >= 0 && wi < p.W)
-                v = p.x[(((long long)st_n * p.H + hi) * p.W + wi) * p.Cin + ci];
+                v = p.x[(((long long)st_n * p.H + hi) * p.W + wi) * p.Cin + cie];
             }
             dst[e] = v;
+            if (++cie == p.Cin) {
+              cie = 0;
+              if (++kwe == p.KW) { kwe = 0; ++khe; }
+            }
           }
         }
       } else {
@@ -224,7 +234,7 @@ __global__ __launch_bounds__(256, TKD == 64 ? 4 : 2) void conv_wgrad_kernel(
     while (st_ho >= p.Ho) { st_ho -= p.Ho; ++st_n; }
   };
 
-  auto write_stage = [&](int buf) {
+  auto write_stage = [&](int buf, unsigned short (&regs)[NPIECE][16]) {
     #pragma unroll
     for (int pi = 0; pi < NPIECE; ++pi) {
       const int sub = st_q + 4 * pi;
@@ -266,15 +276,23 @@ __global__ __launch_bounds__(256, TKD == 64 ? 4 : 2) void conv_wgrad_kernel(
     }
   };
 
-  // ---- main loop over the chunk's m range, register-staged double buffer
+  // ---- main loop: double-buffered LDS, 2-deep register prefetch ----------
+  // stage t+2's loads are issued at iteration t, written to LDS at t+1 —
+  // each load has two full compute phases to land
   const long long n_stages = (m_end - m_begin + 63) >> 6;
-  load_stage();
-  write_stage(0);
+  load_stage(regs0);                           // stage 0
+  write_stage(0, regs0);
+  if (n_stages > 1) load_stage(regs1);         // stage 1
   __syncthreads();
-  for (long long t = 0; t < n_stages; ++t) {
-    if (t + 1 < n_stages) load_stage();
-    compute((int)(t & 1));
-    if (t + 1 < n_stages) write_stage((int)((t + 1) & 1));
+  for (long long t = 0; t < n_stages; ++t) {   // 2x unrolled over reg banks
+    if (t + 2 < n_stages) load_stage(regs0);   // stage t+2
+    compute(0);
+    if (t + 1 < n_stages) write_stage(1, regs1);
+    __syncthreads();
+    if (++t >= n_stages) break;
+    if (t + 2 < n_stages) load_stage(regs1);
+    compute(1);
+    if (t + 1 < n_stages) write_stage(0, regs0);
     __syncthreads();
   }
 
@@ -366,7 +384,9 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   // the combine reduces them in fixed order -> deterministic
   const int tiles = p.kd_tiles * p.co_tiles;
   long long max_chunks = (p.M + 63) / 64;
-  long long want = (768 + tiles - 1) / tiles;
+  int target_blocks = 768;
+  if (const char* e = getenv("IBP_WGRAD_BLOCKS")) target_blocks = atoi(e);
+  long long want = (target_blocks + tiles - 1) / tiles;
   long long chunks = std::min<long long>(std::max<long long>(want, 1), max_chunks);
   p.chunk_len = (((p.M + chunks - 1) / chunks + 63) / 64) * 64;
   p.chunks = (int)((p.M + p.chunk_len - 1) / p.chunk_len);

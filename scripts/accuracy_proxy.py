@@ -1,0 +1,152 @@
+"""End-to-end accuracy proxy: train on rendered synthetic scenes, then score
+recovered keypoints from the FULL process() pipeline against the known joints.
+
+The reference's de-facto correctness check is COCOeval over val2017
+(reference evaluate.py:585-622, AP 0.685) — no dataset exists in this
+environment, so this is the stand-in (VERDICT r1 missing #2): synthetic
+scenes with VISIBLE skeletons (data/synthetic.py render_scene), a short
+training run, and PCK@t over held-out scenes through predict -> find_peaks ->
+find_connections -> find_people.
+
+    python scripts/accuracy_proxy.py --steps 600 --eval 16
+
+Writes the loss curve + PCK table to stdout (redirect into profiles/).
+"""
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+import torch
+
+from improved_body_parts_amd.config import GetConfig, TrainingOpt
+from improved_body_parts_amd.config.inference_params import InferenceParams
+from improved_body_parts_amd.data import SyntheticPoseDataset
+from improved_body_parts_amd.engine import FusedSGD
+from improved_body_parts_amd.engine.inference import (
+    find_connections, find_peaks, find_people, predict)
+from improved_body_parts_amd.models import Network, NetworkEval
+
+
+def pck_score(model_eval, config, ds, indices, thr=0.5, params=None, mp=None):
+    """PCK over held-out scenes: a marked GT joint counts as recovered when a
+    detected peak of the same part type, ASSIGNED TO A PERSON by the greedy
+    assembly, lies within thr * person-scale."""
+    if params is None:
+        params, mp0 = InferenceParams().as_params_dict()
+        mp = mp or dict(mp0)
+        mp["boxsize"] = config.height
+    total = hit = 0
+    n_people_pred = n_people_gt = 0
+    for idx in indices:
+        img, _, _, joints = ds.generate(idx)
+        heat, paf = predict(img, model_eval, config, params, mp)
+        peaks = find_peaks(heat, params, config)
+        conn, special = find_connections(peaks, paf, heat.shape[0], params, config)
+        subset, cand = find_people(conn, special, peaks, params, config)
+        n_people_pred += len(subset)
+        det = [[] for _ in range(config.num_parts)]
+        for s in subset:
+            for part in range(config.num_parts):
+                cid = int(s[part][0])
+                if cid >= 0:
+                    det[part].append(cand[cid][:2])
+        for p in joints:
+            marked = p[:, 2] < 2
+            if not marked.any():
+                continue
+            n_people_gt += 1
+            scale = max(float(np.ptp(p[marked, 1])), 32.0)
+            for part in range(config.num_parts):
+                if p[part, 2] >= 2:
+                    continue
+                total += 1
+                gt = p[part, :2]
+                for d in det[part]:
+                    if np.hypot(d[0] - gt[0], d[1] - gt[1]) <= thr * scale:
+                        hit += 1
+                        break
+    return hit / max(total, 1), n_people_pred, n_people_gt
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=600)
+    ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--nstack", type=int, default=2)
+    ap.add_argument("--lr", type=float, default=7e-4)
+    ap.add_argument("--eval", type=int, default=16)
+    ap.add_argument("--max-people", type=int, default=2)
+    ap.add_argument("--seed", type=int, default=11)
+    args = ap.parse_args()
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda" if use_cuda else "cpu")
+    dtype = torch.bfloat16 if use_cuda else torch.float32
+
+    config = GetConfig("Canonical")
+    opt = TrainingOpt(nstack=args.nstack, batch_size=args.batch,
+                      nstack_weight=[1] * args.nstack)
+    torch.manual_seed(args.seed)
+    model = Network(opt, config, bn=True, dist=True).to(device)
+    if use_cuda:
+        model = model.bfloat16()
+        for m in model.modules():
+            if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+                m.float()
+    model.train()
+    optimizer = FusedSGD(model.parameters(), lr=args.lr, momentum=0.9,
+                         weight_decay=1e-4)
+
+    ds = SyntheticPoseDataset(config, length=1 << 30, seed=args.seed,
+                              max_people=args.max_people, render=True)
+
+    print(f"# accuracy proxy: {args.nstack}-stack IMHN @{config.height}^2, "
+          f"batch {args.batch}, lr {args.lr}, {args.steps} steps, "
+          f"rendered synthetic scenes (<= {args.max_people} people)", flush=True)
+    t0 = time.time()
+    for step in range(args.steps):
+        imgs, mms, hms = [], [], []
+        for b in range(args.batch):
+            img, mm, hm, _ = ds.generate(step * args.batch + b)
+            imgs.append(torch.from_numpy(img))
+            mms.append(torch.from_numpy(mm))
+            hms.append(torch.from_numpy(np.ascontiguousarray(hm)))
+        batch = tuple(torch.stack(t).to(device=device, dtype=dtype)
+                      for t in (imgs, mms, hms))
+        # warmup then flat
+        lr = args.lr * min((step + 1) / 50.0, 1.0)
+        for g in optimizer.param_groups:
+            g["lr"] = lr
+        optimizer.zero_grad(set_to_none=True)
+        loss = model(batch)
+        loss.backward()
+        optimizer.step()
+        if step % 25 == 0 or step == args.steps - 1:
+            print(f"step {step:5d}  loss {float(loss):9.3f}  lr {lr:.2e}  "
+                  f"{time.time() - t0:6.1f}s", flush=True)
+
+    # evaluation through the full pipeline on HELD-OUT scenes
+    model.eval()
+    ev = NetworkEval(opt, config, bn=True).to(device)
+    if use_cuda:
+        ev = ev.bfloat16()
+        for m in ev.modules():
+            if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+                m.float()
+    ev.posenet.load_state_dict(model.posenet.state_dict())
+    ev.eval()
+    held = SyntheticPoseDataset(config, length=1 << 30, seed=args.seed + 999,
+                                max_people=args.max_people, render=True)
+    idx = list(range(args.eval))
+    for thr in (0.5, 0.25):
+        pck, npred, ngt = pck_score(ev, config, held, idx, thr=thr)
+        print(f"PCK@{thr}: {pck:.3f}  (people: predicted {npred} vs GT {ngt}, "
+              f"{args.eval} held-out scenes)", flush=True)
+
+
+if __name__ == "__main__":
+    main()
