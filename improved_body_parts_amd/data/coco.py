@@ -35,26 +35,35 @@ class RawDataIterator:
     (reference py_data_iterator.py:35-144)."""
 
     def __init__(self, global_config, source_config: COCOSourceConfig, shuffle=True,
-                 augment=True):
-        if h5py is None:
+                 augment=True, h5_file=None):
+        """``h5_file``: optional pre-opened mapping with the same two-group
+        layout (``dataset``/``images``/``masks`` of ``[()]``-indexable
+        entries). Lets tests exercise the full read -> convert -> transform ->
+        heatmap path with an in-memory fixture when h5py is unavailable."""
+        if h5py is None and h5_file is None:
             raise RuntimeError("h5py is required for the COCO HDF5 data path") from _H5_ERR
         self.global_config = global_config
         self.source_config = source_config
         self.h5_path = source_config.source()
-        self.h5 = None  # opened lazily per worker process
+        self.h5 = h5_file  # else opened lazily per worker process
         self.keys = None
         self.shuffle = shuffle
         self.augment = augment
         self.heatmapper = Heatmapper(global_config)
         self.transformer = Transformer(global_config)
+        if h5_file is not None:
+            self._bind_groups()
+
+    def _bind_groups(self):
+        self.datum = self.h5["dataset"]
+        self.images = self.h5["images"]
+        self.masks = self.h5.get("masks")
+        self.keys = list(self.datum.keys())
 
     def _ensure_open(self):
         if self.h5 is None:
             self.h5 = h5py.File(self.h5_path, "r")
-            self.datum = self.h5["dataset"]
-            self.images = self.h5["images"]
-            self.masks = self.h5.get("masks")
-            self.keys = list(self.datum.keys())
+            self._bind_groups()
 
     def num_keys(self):
         self._ensure_open()
@@ -93,8 +102,9 @@ class MyDataset(Dataset):
     """torch Dataset adapter (reference data/mydataset.py:15-37)."""
 
     def __init__(self, global_config, config: COCOSourceConfig, shuffle=True,
-                 augment=True):
-        self.iterator = RawDataIterator(global_config, config, shuffle, augment)
+                 augment=True, h5_file=None):
+        self.iterator = RawDataIterator(global_config, config, shuffle, augment,
+                                        h5_file=h5_file)
 
     def __len__(self):
         return self.iterator.num_keys()

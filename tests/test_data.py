@@ -2,7 +2,7 @@ import numpy as np
 import pytest
 import torch
 
-from improved_body_parts_amd.config import CanonicalConfig
+from improved_body_parts_amd.config import CanonicalConfig, COCOSourceConfig, GetConfig
 from improved_body_parts_amd.data import (
     AugmentSelection, SyntheticPoseDataset, Transformer, sample_people,
 )
@@ -123,3 +123,88 @@ def test_transform_image_joint_alignment(cfg):
         window = out_img[int(ny) - 4:int(ny) + 5, int(nx) - 4:int(nx) + 5]
         assert window.max() > 0.2, \
             f"trial {trial}: no bright pixel near transformed joint ({nx},{ny})"
+
+
+# ---------------------------------------------------------------------------
+# COCO h5 data path (in-memory fixture — h5py is absent in this image, so the
+# iterator accepts a pre-opened mapping with the same group layout)
+# ---------------------------------------------------------------------------
+class _FakeDS:
+    def __init__(self, v):
+        self.v = v
+
+    def __getitem__(self, idx):
+        assert idx == ()
+        return self.v
+
+
+class _FakeH5(dict):
+    def get(self, k, default=None):
+        return super().get(k, default)
+
+
+def _coco_fixture(cfg):
+    """One synthetic record in the reference's h5 layout: a 2-person COCO-order
+    annotation, bright dots at joint positions, a mask_miss hole."""
+    import json
+    rng = np.random.default_rng(5)
+    H = W = 480
+    img = (rng.random((H, W, 3)) * 40).astype(np.uint8)
+    src = COCOSourceConfig("unused.h5")
+    people = []
+    for pi in range(2):
+        joints = np.zeros((17, 3), np.float32)
+        cx, cy = 120 + 200 * pi, 180 + 60 * pi
+        for j in range(17):
+            x = cx + (j % 4) * 18
+            y = cy + (j // 4) * 22
+            joints[j] = (x, y, 1.0)
+            img[int(y) - 3:int(y) + 4, int(x) - 3:int(x) + 4] = 255
+        people.append(joints.tolist())
+    mask_miss = np.ones((H, W), np.float32)
+    mask_miss[200:280, 120:260] = 0.0
+    mask_all = np.zeros((H, W), np.float32)
+    mask_all[100:320, 80:380] = 1.0
+    meta = {"image": "img0.jpg", "objpos": [170.0, 230.0],
+            "scale_provided": 120.0 / cfg.height, "joints": people}
+    h5 = _FakeH5({
+        "dataset": {"0": _FakeDS(json.dumps(meta))},
+        "images": {"img0.jpg": _FakeDS(img)},
+        "masks": {"img0.jpg": _FakeDS(np.stack([mask_miss, mask_all]))},
+    })
+    return h5, src
+
+
+def test_coco_h5_iterator_end_to_end():
+    from improved_body_parts_amd.data.coco import MyDataset
+    cfg = GetConfig("Canonical")
+    h5, src = _coco_fixture(cfg)
+    ds = MyDataset(cfg, src, augment=False, h5_file=h5)
+    assert len(ds) == 1
+    img, mask_miss, labels = ds[0]
+    assert img.shape == (cfg.height, cfg.width, 3) and img.dtype == torch.float32
+    assert float(img.max()) <= 1.0 and float(img.max()) > 0.5
+    h, w = cfg.mask_shape
+    assert mask_miss.shape == (1, h, w)
+    assert labels.shape == (cfg.num_layers, h, w)
+    # the mask_miss hole survives the (identity) transform + resize
+    assert float(mask_miss.min()) < 0.5
+    # keypoint channels contain Gaussians where the joints are
+    heat = labels[cfg.heat_start:cfg.bkg_start - 1]
+    assert float(heat.max()) > 0.8
+    # a nose peak lies near the first person's transformed nose position
+    nose = labels[cfg.heat_start]
+    py, px = np.unravel_index(int(nose.argmax()), nose.shape)
+    assert nose.max() > 0.8
+    # paf (limb) channels populated too
+    assert float(labels[:cfg.paf_layers].max()) > 0.5
+
+
+def test_coco_h5_iterator_augmented_runs():
+    from improved_body_parts_amd.data.coco import MyDataset
+    cfg = GetConfig("Canonical")
+    h5, src = _coco_fixture(cfg)
+    ds = MyDataset(cfg, src, augment=True, h5_file=h5)
+    img, mask_miss, labels = ds[0]
+    assert img.shape == (cfg.height, cfg.width, 3)
+    assert torch.isfinite(labels).all()
