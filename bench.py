@@ -72,7 +72,10 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     distributed = world_size > 1
     use_cuda = torch.cuda.is_available()
-    device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
+    # modulo so a 2-rank run on a 1-GPU box (RCCL wiring validation) maps both
+    # ranks onto the existing device instead of an invalid ordinal
+    device = torch.device("cuda", local_rank % max(torch.cuda.device_count(), 1)) \
+        if use_cuda else torch.device("cpu")
     if use_cuda:
         torch.cuda.set_device(device)
     if distributed:
