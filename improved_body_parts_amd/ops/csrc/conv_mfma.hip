@@ -136,8 +136,12 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   const int nk = min(nk_chunk, nk_total - k_begin);
   if (nk <= 0) return;
 
-  unsigned short a_reg[A_ELEMS];
-  unsigned short b_reg[B_ELEMS];
+  // two register banks -> loads are issued TWO stages ahead of their LDS
+  // write (short-K shapes — the 1x1 heads at 4 K-chunks — never reached
+  // steady state with single-stage prefetch). Separate named arrays +
+  // reference parameters keep them in registers.
+  unsigned short a_reg0[A_ELEMS], a_reg1[A_ELEMS];
+  unsigned short b_reg0[B_ELEMS], b_reg1[B_ELEMS];
 
   // ---- incremental tap-walking state for the A gather ----------------------
   // k = (kh*KW + kw)*Cin + ci decodes with ONE division pair at setup; each
@@ -161,7 +165,8 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
 
   // ---- stage the NEXT sequential chunk into registers ----------------------
   // (must be called with consecutive chunks — the tap state walks forward)
-  auto load_chunk = [&](int ck) {
+  auto load_chunk = [&](int ck, unsigned short (&a_reg)[A_ELEMS],
+                        unsigned short (&b_reg)[B_ELEMS]) {
     int kk = t_kk, ci = t_ci, kh = t_kh, kw = t_kw;
     #pragma unroll
     for (int v8 = 0; v8 < A_ELEMS / 8; ++v8) {
@@ -243,7 +248,8 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
     }
   };
 
-  auto write_chunk = [&](int buf) {
+  auto write_chunk = [&](int buf, unsigned short (&a_reg)[A_ELEMS],
+                         unsigned short (&b_reg)[B_ELEMS]) {
     #pragma unroll
     for (int v = 0; v < A_ELEMS / 8; ++v)
       *reinterpret_cast<ushortv8*>(&lds_a[buf][lds_off(a_row, a_off + v * 8)]) =
@@ -278,16 +284,25 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
     }
   };
 
-  // ---- main loop: register-staged double buffer ----------------------------
-  load_chunk(k_begin);
-  write_chunk(0);
+  // ---- main loop: double-buffered LDS, 2-deep register prefetch -----------
+  // chunk t+2's loads are issued at iteration t, written to LDS at t+1 —
+  // each load has two full compute phases to land (2x unrolled so the
+  // register banks stay statically addressed)
+  load_chunk(k_begin, a_reg0, b_reg0);
+  write_chunk(0, a_reg0, b_reg0);
+  if (nk > 1) load_chunk(k_begin + 1, a_reg1, b_reg1);
   __syncthreads();
   for (int t = 0; t < nk; ++t) {
-    if (t + 1 < nk) load_chunk(k_begin + t + 1);  // loads hide under the MFMAs
-    compute(t & 1);
+    if (t + 2 < nk) load_chunk(k_begin + t + 2, a_reg0, b_reg0);
+    compute(0);
     // writing buf[(t+1)&1] is safe without a barrier: its last readers were
     // separated by the end-of-iteration barrier of step t-1
-    if (t + 1 < nk) write_chunk((t + 1) & 1);
+    if (t + 1 < nk) write_chunk(1, a_reg1, b_reg1);
+    __syncthreads();
+    if (++t >= nk) break;
+    if (t + 2 < nk) load_chunk(k_begin + t + 2, a_reg1, b_reg1);
+    compute(1);
+    if (t + 1 < nk) write_chunk(0, a_reg0, b_reg0);
     __syncthreads();
   }
 

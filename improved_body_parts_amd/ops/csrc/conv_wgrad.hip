@@ -72,7 +72,10 @@ __device__ __forceinline__ short4v tr16_read(const unsigned short* p) {
       (__attribute__((address_space(3))) short4v*)p);
 }
 
-template <int TKD, int TCO>
+// ELEM compiles in the tap-crossing per-element path (stem, Cin % 16 != 0,
+// partial last kd-tile) — its precomputed pe_pack walk costs ~16 VGPRs, so
+// the clean-tiled hot layers instantiate ELEM=false and keep occupancy 4
+template <int TKD, int TCO, bool ELEM>
 __global__ __launch_bounds__(256, TKD == 64 ? 3 : 2) void conv_wgrad_kernel(
     WgradParams p) {
   constexpr int ASUB = TKD / 16;
@@ -121,18 +124,44 @@ __global__ __launch_bounds__(256, TKD == 64 ? 3 : 2) void conv_wgrad_kernel(
   bool pc_elem[NPIECE];     // piece crosses a tap boundary -> per-element path
   bool pc_isA[NPIECE];
   int pc_co[NPIECE];
+  // tap-crossing pieces (stem Cin=3, Cin % 16 != 0 tails): the per-element
+  // (ci, kh, kw) walk is STAGE-INVARIANT — precompute it once, packed as
+  // ci | kh<<10 | kw<<20 | dead<<30, and decode with shifts in the loop
+  // (round-5 measurement: a non-unrolled runtime walk per stage cost 2-3x;
+  // an unrolled division chain spilled registers)
+  int pe_pack[ELEM ? 16 : 1];
+  if (ELEM) {
+    #pragma unroll
+    for (int e = 0; e < (ELEM ? 16 : 1); ++e) pe_pack[e] = 1 << 30;
+  }
   #pragma unroll
   for (int pi = 0; pi < NPIECE; ++pi) {
     const int sub = st_q + 4 * pi;
     if (sub < ASUB) {
       const int k0 = kd0 + 16 * sub;
-      const int f = k0 / p.Cin;
+      const int f = p.Cin > 0 ? k0 / p.Cin : 0;
       pc_isA[pi] = true;
       pc_ci[pi] = k0 - f * p.Cin;
       pc_kh[pi] = f / p.KW;
       pc_kw[pi] = f - pc_kh[pi] * p.KW;
-      pc_elem[pi] = (pc_ci[pi] + 16 > p.Cin) || (k0 + 16 > p.KD);
+      pc_elem[pi] = ELEM &&
+                    ((pc_ci[pi] + 16 > p.Cin) || (k0 + 16 > p.KD));
       pc_co[pi] = 0;
+      if (ELEM && pc_elem[pi]) {
+        // at most one A piece per thread (ASUB == 4, st_q in 0..3; the
+        // 128-tile has two A pieces per thread but is only selected for
+        // clean-tiled shapes where pc_elem is impossible)
+        int cie = pc_ci[pi], khe = pc_kh[pi], kwe = pc_kw[pi];
+        #pragma unroll 1
+        for (int e = 0; e < 16; ++e) {
+          pe_pack[ELEM ? e : 0] =
+              (k0 + e < p.KD) ? (cie | (khe << 10) | (kwe << 20)) : (1 << 30);
+          if (++cie == p.Cin) {
+            cie = 0;
+            if (++kwe == p.KW) { kwe = 0; ++khe; }
+          }
+        }
+      }
     } else {
       pc_isA[pi] = false;
       pc_co[pi] = co0 + 16 * (sub - ASUB);
@@ -187,26 +216,26 @@ __global__ __launch_bounds__(256, TKD == 64 ? 3 : 2) void conv_wgrad_kernel(
             #pragma unroll
             for (int e = 0; e < 16; ++e) dst[e] = 0;
           }
-        } else {
-          // tap-crossing piece (stem, Cin % 16 != 0): walk (ci, kh, kw)
-          // element by element — division-free, NOT unrolled (an unrolled
-          // 16-wide division chain ballooned VGPR pressure into spills)
-          const int k0 = kd0 + 16 * (st_q + 4 * pi);
-          int cie = pc_ci[pi], khe = pc_kh[pi], kwe = pc_kw[pi];
-          #pragma unroll 1
+        } else if (ELEM) {
+          // tap-crossing piece: decode the precomputed packed walk — shifts
+          // and adds only, fully unrolled for ILP
+          const int hb = st_ho * p.stride - p.pad_h;
+          const int wb = st_wo * p.stride - p.pad_w;
+          const long long nbase = (long long)st_n * p.H;
+          #pragma unroll
           for (int e = 0; e < 16; ++e) {
+            const int pk = pe_pack[ELEM ? e : 0];
             unsigned short v = 0;
-            if (m_ok && k0 + e < p.KD) {
-              const int hi = st_ho * p.stride - p.pad_h + khe * p.dil_h;
-              const int wi = st_wo * p.stride - p.pad_w + kwe * p.dil_w;
+            if (m_ok && pk < (1 << 30)) {
+              const int cie = pk & 1023;
+              const int khe = (pk >> 10) & 1023;
+              const int kwe = pk >> 20;
+              const int hi = hb + khe * p.dil_h;
+              const int wi = wb + kwe * p.dil_w;
               if (hi >= 0 && hi < p.H && wi >= 0 && wi < p.W)
-                v = p.x[(((long long)st_n * p.H + hi) * p.W + wi) * p.Cin + cie];
+                v = p.x[((nbase + hi) * p.W + wi) * p.Cin + cie];
             }
             dst[e] = v;
-            if (++cie == p.Cin) {
-              cie = 0;
-              if (++kwe == p.KW) { kwe = 0; ++khe; }
-            }
           }
         }
       } else {
@@ -398,11 +427,20 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   // every in-range (kd, cout) is written by exactly one block per chunk;
   // out-of-range rows are never read back -> no zero-init needed
   dim3 grid(p.chunks * tiles), block(256);
-  if (big) {
-    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128>), grid, block, 0,
+  // tap-crossing pieces exist iff Cin is not 16-aligned or the last kd-tile
+  // is partial — only then compile in the per-element path (costs VGPRs)
+  const bool elem = (Cin % 16 != 0) || (p.KD % (big ? 128 : 64) != 0);
+  if (big && elem) {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128, true>), grid, block,
+                       0, stream, p);
+  } else if (big) {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128, false>), grid, block,
+                       0, stream, p);
+  } else if (elem) {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<64, 64, true>), grid, block, 0,
                        stream, p);
   } else {
-    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<64, 64>), grid, block, 0,
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<64, 64, false>), grid, block, 0,
                        stream, p);
   }
 
