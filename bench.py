@@ -266,6 +266,8 @@ def main():
                 "global_batch": batch * world_size,
                 "input": args.input,
                 "parallelism": f"dp{world_size}",
+                "max_mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 2)
+                if use_cuda else None,
             },
         }), flush=True)
 

@@ -32,14 +32,17 @@ from improved_body_parts_amd.models import Network, NetworkEval
 
 
 def pck_score(model_eval, config, ds, indices, thr=0.5, params=None, mp=None):
-    """PCK over held-out scenes: a marked GT joint counts as recovered when a
-    detected peak of the same part type, ASSIGNED TO A PERSON by the greedy
-    assembly, lies within thr * person-scale."""
+    """PCK (recall) over held-out scenes: a marked GT joint counts as
+    recovered when a detected peak of the same part type, ASSIGNED TO A
+    PERSON by the greedy assembly, lies within thr * person-scale. Also
+    returns joint precision: the fraction of assigned detections that match
+    some GT joint of their type at the same radius."""
     if params is None:
         params, mp0 = InferenceParams().as_params_dict()
         mp = mp or dict(mp0)
         mp["boxsize"] = config.height
     total = hit = 0
+    det_total = det_matched = 0
     n_people_pred = n_people_gt = 0
     for idx in indices:
         img, _, _, joints = ds.generate(idx)
@@ -54,6 +57,7 @@ def pck_score(model_eval, config, ds, indices, thr=0.5, params=None, mp=None):
                 cid = int(s[part][0])
                 if cid >= 0:
                     det[part].append(cand[cid][:2])
+        gt_by_part = [[] for _ in range(config.num_parts)]
         for p in joints:
             marked = p[:, 2] < 2
             if not marked.any():
@@ -63,13 +67,21 @@ def pck_score(model_eval, config, ds, indices, thr=0.5, params=None, mp=None):
             for part in range(config.num_parts):
                 if p[part, 2] >= 2:
                     continue
+                gt_by_part[part].append((p[part, 0], p[part, 1], scale))
                 total += 1
                 gt = p[part, :2]
                 for d in det[part]:
                     if np.hypot(d[0] - gt[0], d[1] - gt[1]) <= thr * scale:
                         hit += 1
                         break
-    return hit / max(total, 1), n_people_pred, n_people_gt
+        for part in range(config.num_parts):
+            for d in det[part]:
+                det_total += 1
+                if any(np.hypot(d[0] - gx, d[1] - gy) <= thr * gs
+                       for gx, gy, gs in gt_by_part[part]):
+                    det_matched += 1
+    return (hit / max(total, 1), det_matched / max(det_total, 1),
+            n_people_pred, n_people_gt)
 
 
 def main():
@@ -143,8 +155,9 @@ def main():
                                 max_people=args.max_people, render=True)
     idx = list(range(args.eval))
     for thr in (0.5, 0.25):
-        pck, npred, ngt = pck_score(ev, config, held, idx, thr=thr)
-        print(f"PCK@{thr}: {pck:.3f}  (people: predicted {npred} vs GT {ngt}, "
+        pck, prec, npred, ngt = pck_score(ev, config, held, idx, thr=thr)
+        print(f"PCK@{thr}: {pck:.3f}  joint-precision {prec:.3f}  "
+              f"(people: predicted {npred} vs GT {ngt}, "
               f"{args.eval} held-out scenes)", flush=True)
 
 
