@@ -136,12 +136,12 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   const int nk = min(nk_chunk, nk_total - k_begin);
   if (nk <= 0) return;
 
-  // two register banks -> loads are issued TWO stages ahead of their LDS
-  // write (short-K shapes — the 1x1 heads at 4 K-chunks — never reached
-  // steady state with single-stage prefetch). Separate named arrays +
-  // reference parameters keep them in registers.
-  unsigned short a_reg0[A_ELEMS], a_reg1[A_ELEMS];
-  unsigned short b_reg0[B_ELEMS], b_reg1[B_ELEMS];
+  // single-stage register prefetch: a 2-deep variant (as in conv_wgrad.hip)
+  // was measured 10-15% SLOWER here — the fwd gather stages 2-4x more
+  // register bytes per thread than wgrad, and the doubled banks cost more in
+  // scheduling/pressure than the extra latency cover buys
+  unsigned short a_reg0[A_ELEMS];
+  unsigned short b_reg0[B_ELEMS];
 
   // ---- incremental tap-walking state for the A gather ----------------------
   // k = (kh*KW + kw)*Cin + ci decodes with ONE division pair at setup; each
@@ -284,25 +284,16 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
     }
   };
 
-  // ---- main loop: double-buffered LDS, 2-deep register prefetch -----------
-  // chunk t+2's loads are issued at iteration t, written to LDS at t+1 —
-  // each load has two full compute phases to land (2x unrolled so the
-  // register banks stay statically addressed)
+  // ---- main loop: register-staged double buffer ----------------------------
   load_chunk(k_begin, a_reg0, b_reg0);
   write_chunk(0, a_reg0, b_reg0);
-  if (nk > 1) load_chunk(k_begin + 1, a_reg1, b_reg1);
   __syncthreads();
   for (int t = 0; t < nk; ++t) {
-    if (t + 2 < nk) load_chunk(k_begin + t + 2, a_reg0, b_reg0);
-    compute(0);
+    if (t + 1 < nk) load_chunk(k_begin + t + 1, a_reg0, b_reg0);
+    compute(t & 1);
     // writing buf[(t+1)&1] is safe without a barrier: its last readers were
     // separated by the end-of-iteration barrier of step t-1
-    if (t + 1 < nk) write_chunk(1, a_reg1, b_reg1);
-    __syncthreads();
-    if (++t >= nk) break;
-    if (t + 2 < nk) load_chunk(k_begin + t + 2, a_reg1, b_reg1);
-    compute(1);
-    if (t + 1 < nk) write_chunk(0, a_reg0, b_reg0);
+    if (t + 1 < nk) write_chunk((t + 1) & 1, a_reg0, b_reg0);
     __syncthreads();
   }
 
