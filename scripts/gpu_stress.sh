@@ -7,12 +7,12 @@ set -x
 mkdir -p gpurun_out
 for B in 4 16 32 48; do
   timeout 420 python bench.py --mode train --input 768 --nstack 8 --batch $B \
-    --steps 10 --warmup 3 > gpurun_out/stress_b$B.log 2>&1
+    --steps 6 --warmup 2 > gpurun_out/stress_b$B.log 2>&1
   echo "batch $B rc=$?:"
   tail -1 gpurun_out/stress_b$B.log
 done
 timeout 420 python bench.py --mode train --input 768 --nstack 8 --batch 16 \
-  --data device-gt --steps 10 --warmup 3 > gpurun_out/stress_devgt.log 2>&1
+  --data device-gt --steps 6 --warmup 2 > gpurun_out/stress_devgt.log 2>&1
 echo "device-gt rc=$?:"
 tail -1 gpurun_out/stress_devgt.log
 rocm-smi --showmeminfo vram | tail -5
