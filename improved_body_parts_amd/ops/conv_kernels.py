@@ -72,6 +72,100 @@ def packed_weight_dgrad(weight: torch.Tensor) -> torch.Tensor:
                         .permute(1, 2, 3, 0).reshape(w.shape[1], -1).contiguous())
 
 
+# ---------------------------------------------------------------------------
+# 7x7 s2 stem as a space-to-depth 4x4 s1 conv
+# ---------------------------------------------------------------------------
+# The Cin=3 stem forces the kernel's per-element tap gather (Cin % 8 != 0).
+# Rewriting x (N,3,H,W) as S (N,12->16,H/2,W/2) with channel
+# c' = (py*2+px)*3 + c turns it into a 4x4 stride-1 conv over 16 channels
+# (kh = 2*kh' + py - 1, effective pad (2 left, 1 right)) — every subpiece is
+# tap-uniform and 16-B vectorizable. Same trick for wgrad; dgrad is never
+# needed (the stem is the input layer).
+
+def _is_stem(x, weight, stride, padding, dilation):
+    return (weight.shape[2] == 7 and weight.shape[3] == 7
+            and stride[0] == 2 and stride[1] == 2 and x.shape[1] == 3
+            and padding[0] == 3 and padding[1] == 3
+            and dilation[0] == 1 and dilation[1] == 1
+            and x.shape[2] % 2 == 0 and x.shape[3] % 2 == 0)
+
+
+def _stem_s2d_input(x):
+    """(N,3,H,W) channels_last -> (N,16,H/2,W/2) channels_last (padded c')."""
+    import torch.nn.functional as F
+    n, _, h, w = x.shape
+    v = x.contiguous(memory_format=_CL).permute(0, 2, 3, 1)  # N,H,W,3
+    v = v.reshape(n, h // 2, 2, w // 2, 2, 3).permute(0, 1, 3, 2, 4, 5)
+    v = v.reshape(n, h // 2, w // 2, 12)
+    v = F.pad(v, (0, 4))                                     # c' 12..15 = 0
+    return v.permute(0, 3, 1, 2)  # NCHW view over NHWC storage
+
+
+_STEM_LUT = {}
+
+
+def _stem_weight_lut(device):
+    """Flat source indices: W'[c'][kh'][kw'] position backing W[c][kh][kw]."""
+    lut = _STEM_LUT.get(device)
+    if lut is not None:
+        return lut
+    import numpy as np
+    src = np.zeros(3 * 7 * 7, np.int64)
+    for c in range(3):
+        for kh in range(7):
+            for kw in range(7):
+                py, kh2 = (kh + 1) & 1, (kh + 1) >> 1
+                px, kw2 = (kw + 1) & 1, (kw + 1) >> 1
+                cp = (py * 2 + px) * 3 + c
+                src[(c * 7 + kh) * 7 + kw] = (cp * 4 + kh2) * 4 + kw2
+    lut = torch.from_numpy(src).to(device)
+    _STEM_LUT[device] = lut
+    return lut
+
+
+def _stem_pack_weight(weight):
+    """[64,3,7,7] -> packed bf16 [64][K=4*4*16] for the s2d conv."""
+    def pack(w):
+        cout = w.shape[0]
+        wp = torch.zeros(cout, 16, 4, 4, dtype=w.dtype, device=w.device)
+        lut = _stem_weight_lut(w.device)
+        wp.view(cout, -1)[:, lut] = w.reshape(cout, -1)
+        # -> [cout][kh'*kw'*c'] (tap-major, channel fastest) like packed_weight
+        return wp.permute(0, 2, 3, 1).reshape(cout, -1).contiguous()
+    return _cached_pack(weight, "s2d", pack)
+
+
+def _stem_fwd(x, weight, scale, shift, residual, act,
+              residual_post, residual_post2):
+    ext = hip_extension()
+    s = _stem_s2d_input(x).contiguous(memory_format=_CL)
+    n, _, hs, ws = s.shape
+    cout = weight.shape[0]
+    if residual is not None:
+        residual = residual.contiguous(memory_format=_CL)
+    if residual_post is not None:
+        residual_post = residual_post.contiguous(memory_format=_CL)
+    if residual_post2 is not None:
+        residual_post2 = residual_post2.contiguous(memory_format=_CL)
+    y = ext.conv_mfma_fwd(s, _stem_pack_weight(weight), n, hs, ws, 16, cout,
+                          4, 4, 1, 2, 2, 1, 1, hs, ws,
+                          scale, shift, residual, act,
+                          residual_post, residual_post2, 1)
+    return y.permute(0, 3, 1, 2)
+
+
+def _stem_wgrad(x, dy, w_shape):
+    ext = hip_extension()
+    s = _stem_s2d_input(x).contiguous(memory_format=_CL)
+    dy = dy.contiguous(memory_format=_CL)
+    n, _, hs, ws = s.shape
+    cout = w_shape[0]
+    dwp = ext.conv_mfma_wgrad(s, dy, n, hs, ws, 16, cout, 4, 4,
+                              1, 2, 2, 1, 1, dy.shape[2], dy.shape[3])
+    lut = _stem_weight_lut(x.device)
+    return dwp.reshape(cout, -1)[:, lut].reshape(cout, 3, 7, 7)
+
+
 def conv_fwd(x, weight, stride, padding, dilation,
              scale=None, shift=None, residual=None, act=False,
              residual_post=None, residual_post2=None):
@@ -85,6 +179,9 @@ def conv_fwd(x, weight, stride, padding, dilation,
     ext = hip_extension()
     if not hasattr(ext, "conv_mfma_fwd"):
         return None
+    if _is_stem(x, weight, stride, padding, dilation):
+        return _stem_fwd(x, weight, scale, shift, residual, act,
+                         residual_post, residual_post2)
     x = x.contiguous(memory_format=_CL)
     n, cin, h, w_ = x.shape
     cout, _, kh, kw = weight.shape
@@ -143,6 +240,10 @@ def conv_wgrad(x, dy, w_shape, stride, padding, dilation):
     ext = hip_extension()
     if not hasattr(ext, "conv_mfma_wgrad"):
         return None
+    if w_shape[2] == 7 and w_shape[3] == 7 and x.shape[1] == 3 \
+            and stride[0] == 2 and padding[0] == 3 and dilation[0] == 1 \
+            and x.shape[2] % 2 == 0 and x.shape[3] % 2 == 0:
+        return _stem_wgrad(x, dy, w_shape)
     x = x.contiguous(memory_format=_CL)
     dy = dy.contiguous(memory_format=_CL)
     n, cin, h, w_ = x.shape

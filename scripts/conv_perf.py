@@ -21,6 +21,7 @@ shapes = [  # (n, cin, cout, hw, k, s, d) — hottest IMHN layers at batch 4 & 1
     (16, 768, 384, 8, 1, 1, 1),
     (16, 384, 384, 8, 3, 1, 1),
     (4, 256, 256, 128, 3, 1, 1),
+    (16, 3, 64, 512, 7, 2, 1),   # the stem (space-to-depth path)
 ]
 for (n, cin, cout, hw, k, s, d) in shapes:
     x = torch.randn(n, cin, hw, hw, device="cuda").bfloat16().contiguous(memory_format=CL)
