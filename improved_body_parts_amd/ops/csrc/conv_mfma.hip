@@ -168,6 +168,31 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   auto load_chunk = [&](int ck, unsigned short (&a_reg)[A_ELEMS],
                         unsigned short (&b_reg)[B_ELEMS]) {
     int kk = t_kk, ci = t_ci, kh = t_kh, kw = t_kw;
+    // fast path: the whole A_ELEMS piece inside one filter tap (all the
+    // Cin % 64 == 0 hourglass convs) — ONE predicate set, straight-line
+    // vector loads (the per-subpiece form below costs ~10% on hot shapes)
+    if (ci + A_ELEMS <= p.Cin && kk + A_ELEMS <= p.K && p.zs == 1) {
+      const int hi = hi_base + kh * p.dil_h;
+      const int wi = wi_base + kw * p.dil_w;
+      const bool ok = a_valid_row && hi >= 0 && hi < p.H &&
+                      wi >= 0 && wi < p.W;
+      const unsigned short* src =
+          p.x + (((long long)a_n * p.H + hi) * p.W + wi) * p.Cin + ci;
+      const bool aligned = ((reinterpret_cast<uintptr_t>(src)) & 15u) == 0;
+      if (ok && aligned) {
+        #pragma unroll
+        for (int v = 0; v < A_ELEMS / 8; ++v)
+          *reinterpret_cast<ushortv8*>(&a_reg[v * 8]) =
+              *reinterpret_cast<const ushortv8*>(src + v * 8);
+      } else if (ok) {
+        #pragma unroll
+        for (int e = 0; e < A_ELEMS; ++e) a_reg[e] = src[e];
+      } else {
+        #pragma unroll
+        for (int e = 0; e < A_ELEMS; ++e) a_reg[e] = 0;
+      }
+      goto a_done;
+    }
     #pragma unroll
     for (int v8 = 0; v8 < A_ELEMS / 8; ++v8) {
       unsigned short* dst = &a_reg[v8 * 8];
@@ -225,6 +250,7 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
       kk += 8;
       tap_advance(ci, kh, kw, 8);
     }
+a_done:
     // walk the persistent state one full chunk forward
     t_kk += BK;
     tap_advance(t_ci, t_kh, t_kw, BK);

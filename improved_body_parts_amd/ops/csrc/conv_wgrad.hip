@@ -76,7 +76,8 @@ __device__ __forceinline__ short4v tr16_read(const unsigned short* p) {
 // partial last kd-tile) — its precomputed pe_pack walk costs ~16 VGPRs, so
 // the clean-tiled hot layers instantiate ELEM=false and keep occupancy 4
 template <int TKD, int TCO, bool ELEM>
-__global__ __launch_bounds__(256, TKD == 64 ? 3 : 2) void conv_wgrad_kernel(
+__global__
+__launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
     WgradParams p) {
   constexpr int ASUB = TKD / 16;
   constexpr int BSUB = TCO / 16;
@@ -345,6 +346,26 @@ __global__ __launch_bounds__(256, TKD == 64 ? 3 : 2) void conv_wgrad_kernel(
   }
 }
 
+// tree stage: partial[g][e] = sum of ws[g*G .. g*G+G)[e] — full element x
+// group parallelism (a flat chunk loop was latency-bound on the 1x1 layers,
+// whose tiny tile grids need ~100-200 M-chunks: 6% of the train step)
+__global__ void wgrad_reduce_stage_kernel(const float* __restrict__ ws,
+                                          float* __restrict__ out,
+                                          long long per_chunk, int chunks,
+                                          int G, int groups) {
+  const long long total = per_chunk * groups;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int g = (int)(i / per_chunk);
+    const long long e = i - (long long)g * per_chunk;
+    const int c0 = g * G;
+    const int c1 = min(c0 + G, chunks);
+    float v = 0.f;
+    for (int c = c0; c < c1; ++c) v += ws[(long long)c * per_chunk + e];
+    out[i] = v;
+  }
+}
+
 // combine: dW[cout][cin][kh][kw] (bf16) = sum over chunks of ws[c][f*Cin+ci][cout]
 __global__ void wgrad_combine_kernel(const float* __restrict__ ws,
                                      unsigned short* __restrict__ dw,
@@ -400,12 +421,18 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   p.KD = (int)(KH * KW * Cin);
   p.step_dho = 64 / p.Wo;
   p.step_dwo = 64 % p.Wo;
-  // 64x64 tiles at 4 blocks/CU measured faster than 128x128 at 2 blocks/CU
-  // on every shape (the kernel is load-latency-, not LDS-, bound); the big
-  // variant is kept behind an env knob for tile experiments
+  // Tile selection. The wide-spatial 3x3 layers are L3-BANDWIDTH-bound on
+  // re-reads: with a TCOxTKD grid each x byte is read (taps * Cout/TCO)
+  // times and each dy byte (KD/TKD) times per chunk (~9.7 GB for the
+  // 256ch@128^2 layer at 64x64 — about the measured runtime at L3 rate).
+  // TCO=128 halves both factors; TKD stays 64 (128x128 at 2 blocks/CU
+  // measured slower — latency). Small-M layers keep 64x64 (fit in L2, and
+  // the 12-subtile LDS drops occupancy 4 -> 3).
   const char* bt = getenv("IBP_WGRAD_BIG");
   const bool big = bt && bt[0] == '1' && p.KD >= 128 && Cout >= 128;
-  const int TKD = big ? 128 : 64, TCO = big ? 128 : 64;
+  const bool wide = !big && Cout >= 128 && p.M >= 65536;
+  const int TKD = big ? 128 : 64;
+  const int TCO = big ? 128 : (wide ? 128 : 64);
   p.kd_tiles = (p.KD + TKD - 1) / TKD;
   p.co_tiles = (int)((Cout + TCO - 1) / TCO);
   // split M into chunks for parallelism: aim ~768 blocks, chunk length a
@@ -429,12 +456,18 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   dim3 grid(p.chunks * tiles), block(256);
   // tap-crossing pieces exist iff Cin is not 16-aligned or the last kd-tile
   // is partial — only then compile in the per-element path (costs VGPRs)
-  const bool elem = (Cin % 16 != 0) || (p.KD % (big ? 128 : 64) != 0);
+  const bool elem = (Cin % 16 != 0) || (p.KD % TKD != 0);
   if (big && elem) {
     hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128, true>), grid, block,
                        0, stream, p);
   } else if (big) {
     hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128, false>), grid, block,
+                       0, stream, p);
+  } else if (wide && elem) {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<64, 128, true>), grid, block, 0,
+                       stream, p);
+  } else if (wide) {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<64, 128, false>), grid, block,
                        0, stream, p);
   } else if (elem) {
     hipLaunchKernelGGL((ibp::conv_wgrad_kernel<64, 64, true>), grid, block, 0,
@@ -446,11 +479,28 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
 
   Tensor dw = torch::empty({Cout, Cin, KH, KW}, x.options());
   long long total = Cout * Cin * KH * KW;
+  const float* ws_ptr = p.ws;
+  int chunks_left = p.chunks;
+  Tensor partial;
+  if (chunks_left > 16) {
+    // tree stage so the final scatter loops over <= 16 slices
+    const int G = 8;
+    const int groups = (chunks_left + G - 1) / G;
+    const long long per_chunk = (long long)p.KD * Cout;
+    partial = torch::empty({(long long)groups * per_chunk},
+                           x.options().dtype(torch::kFloat32));
+    dim3 rgrid(ibp::grid_1d(per_chunk * groups, 256, 8192)), rblock(256);
+    hipLaunchKernelGGL(ibp::wgrad_reduce_stage_kernel, rgrid, rblock, 0,
+                       stream, ws_ptr, partial.data_ptr<float>(), per_chunk,
+                       chunks_left, G, groups);
+    ws_ptr = partial.data_ptr<float>();
+    chunks_left = groups;
+  }
   dim3 cgrid(ibp::grid_1d(total, 256, 4096)), cblock(256);
   hipLaunchKernelGGL(ibp::wgrad_combine_kernel, cgrid, cblock, 0, stream,
-                     p.ws, reinterpret_cast<unsigned short*>(dw.data_ptr()),
+                     ws_ptr, reinterpret_cast<unsigned short*>(dw.data_ptr()),
                      total, (int)Cin, (int)(KH * KW), (int)Cout, p.KD,
-                     p.chunks);
+                     chunks_left);
   return dw;
 }
 
