@@ -430,7 +430,11 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   // the 12-subtile LDS drops occupancy 4 -> 3).
   const char* bt = getenv("IBP_WGRAD_BIG");
   const bool big = bt && bt[0] == '1' && p.KD >= 128 && Cout >= 128;
-  const bool wide = !big && Cout >= 128 && p.M >= 65536;
+  // wide tile measured: 1x1 @128^2 0.78 -> 0.87x MIOpen, but 3x3 0.60 ->
+  // 0.37x (the occupancy-3->2 latency cost beats the halved re-reads when
+  // the tap factor, not Cout/TCO, dominates traffic) -> 1x1 only
+  const bool wide = !big && KH == 1 && KW == 1 && Cout >= 128 &&
+                    p.M >= 65536;
   const int TKD = big ? 128 : 64;
   const int TCO = big ? 128 : (wide ? 128 : 64);
   p.kd_tiles = (p.KD + TKD - 1) / TKD;
