@@ -446,9 +446,11 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
   // sums their UNINITIALISED workspace slices (silent garbage whenever the
   // allocator hands back dirty memory — found as exploding gradients through
   // the scale-2 Merge convs, scripts/diag_splitk.py).
+  int sk_target = 384;
+  if (const char* e = getenv("IBP_SPLITK_TARGET")) sk_target = atoi(e);
   int splitk = 1;
-  if (ntiles < 384 && nk_total > 1) {
-    splitk = std::min((int)nk_total, (384 + ntiles - 1) / ntiles);
+  if (ntiles < sk_target && nk_total > 1) {
+    splitk = std::min((int)nk_total, (sk_target + ntiles - 1) / ntiles);
     int nk_chunk = ((int)nk_total + splitk - 1) / splitk;
     splitk = ((int)nk_total + nk_chunk - 1) / nk_chunk;
   }

@@ -44,3 +44,25 @@ for k, d in sorted(agg.items(), key=lambda kv: -kv[1].get('SQ_WAVE_CYCLES', 0))[
           f"valu_insts={d.get('SQ_INSTS_VALU',0):.3e} lds_conf={d.get('SQ_LDS_BANK_CONFLICT',0):.3e}")
 PYEOF
 cat $R/gpurun_out/wgrad_pmc.txt
+
+# second pass: memory counters (separate run — cannot combine with traces)
+timeout 600 rocprofv3 --pmc FETCH_SIZE WRITE_SIZE SQ_WAVE_CYCLES SQ_VALU_MFMA_BUSY_CYCLES SQ_WAIT_ANY --output-format csv -d /tmp/wprof2 -o wpmc2 -- python /tmp/wgrad_pmc_driver.py > /tmp/wpmc2.log 2>&1
+echo "pmc2=$?"
+python - <<'PYEOF' >> $R/gpurun_out/wgrad_pmc.txt 2>&1
+import csv, glob, collections
+f = sorted(glob.glob('/tmp/wprof2/**/*counter_collection.csv', recursive=True))
+print('=== memory counters ===')
+agg = collections.defaultdict(lambda: collections.defaultdict(float))
+for fn in f:
+    with open(fn) as fh:
+        for row in csv.DictReader(fh):
+            k = row.get('Kernel_Name', row.get('kernel_name', '?'))[:70]
+            c = row.get('Counter_Name', row.get('counter_name', '?'))
+            agg[k][c] += float(row.get('Counter_Value', row.get('counter_value', 0)))
+for k, d in sorted(agg.items(), key=lambda kv: -kv[1].get('SQ_WAVE_CYCLES', 0))[:6]:
+    wc = d.get('SQ_WAVE_CYCLES', 1)
+    print(k)
+    print(f"   fetch={d.get('FETCH_SIZE',0)/1e6:.1f} MB write={d.get('WRITE_SIZE',0)/1e6:.1f} MB "
+          f"mfma={d.get('SQ_VALU_MFMA_BUSY_CYCLES',0)/wc*100:5.1f}% wait={d.get('SQ_WAIT_ANY',0)/wc*100:5.1f}%")
+PYEOF
+tail -16 $R/gpurun_out/wgrad_pmc.txt
