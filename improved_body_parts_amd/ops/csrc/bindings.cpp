@@ -60,6 +60,9 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
                      const c10::optional<Tensor>& residual_post,
                      const c10::optional<Tensor>& residual_post2, int64_t zs);
 
+void pack_conv_weight(const Tensor& w, Tensor& fwd_pack,
+                      const c10::optional<Tensor>& dgrad_pack);
+
 // conv_wgrad.hip
 Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
                        int64_t W, int64_t Cin, int64_t Cout, int64_t KH,
@@ -112,6 +115,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("residual") = c10::nullopt, py::arg("act") = false,
         py::arg("residual_post") = c10::nullopt,
         py::arg("residual_post2") = c10::nullopt, py::arg("zs") = 1);
+  m.def("pack_conv_weight", &pack_conv_weight,
+        "fwd + dgrad weight packs in one kernel",
+        py::arg("w"), py::arg("fwd_pack"), py::arg("dgrad_pack") = c10::nullopt);
   m.def("conv_mfma_wgrad", &conv_mfma_wgrad,
         "MFMA implicit-GEMM conv weight gradient (tr16 LDS transpose)");
   m.def("tr16_probe", &tr16_probe,

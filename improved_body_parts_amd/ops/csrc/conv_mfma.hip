@@ -380,10 +380,55 @@ __global__ void splitk_combine_kernel(const float* __restrict__ ws,
   }
 }
 
+// weight pack: [Cout,Cin,KH,KW] (standard contiguous) -> fwd [Cout][f*Cin+ci]
+// and (optional) dgrad [Cin][(rot180 f)*Cout+co] in ONE pass. Weights change
+// every optimizer step, so the eager flip/permute/reshape chains re-ran per
+// layer per step (~500 small launches); this is one launch per weight.
+__global__ void pack_weight_kernel(const unsigned short* __restrict__ w,
+                                   unsigned short* __restrict__ fwd,
+                                   unsigned short* __restrict__ dgr,
+                                   int Cout, int Cin, int KH, int KW) {
+  const long long total = (long long)Cout * Cin * KH * KW;
+  const int KHW = KH * KW;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int kw = (int)(i % KW);
+    long long t = i / KW;
+    const int kh = (int)(t % KH); t /= KH;
+    const int ci = (int)(t % Cin);
+    const int co = (int)(t / Cin);
+    const unsigned short v = w[i];
+    const int f = kh * KW + kw;
+    fwd[(long long)co * KHW * Cin + f * Cin + ci] = v;
+    if (dgr) {
+      const int fr = (KH - 1 - kh) * KW + (KW - 1 - kw);
+      dgr[(long long)ci * KHW * Cout + fr * Cout + co] = v;
+    }
+  }
+}
+
 }  // namespace ibp
 
 // ===========================================================================
 using torch::Tensor;
+
+void pack_conv_weight(const Tensor& w, Tensor& fwd_pack,
+                      const c10::optional<Tensor>& dgrad_pack) {
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous() &&
+              w.scalar_type() == at::ScalarType::BFloat16);
+  int Cout = (int)w.size(0), Cin = (int)w.size(1);
+  int KH = (int)w.size(2), KW = (int)w.size(3);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  dim3 grid(ibp::grid_1d(w.numel(), 256, 2048)), block(256);
+  hipLaunchKernelGGL(
+      ibp::pack_weight_kernel, grid, block, 0, stream,
+      reinterpret_cast<const unsigned short*>(w.data_ptr()),
+      reinterpret_cast<unsigned short*>(fwd_pack.data_ptr()),
+      dgrad_pack.has_value()
+          ? reinterpret_cast<unsigned short*>(dgrad_pack->data_ptr())
+          : nullptr,
+      Cout, Cin, KH, KW);
+}
 
 Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
                      int64_t H, int64_t W, int64_t Cin, int64_t Cout,

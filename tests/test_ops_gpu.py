@@ -220,6 +220,20 @@ def test_conv_post_act_residuals(training):
         _assert_rel(r2g.grad, dy_ref, 2e-2, "dres_post2 identity")
 
 
+def test_pack_conv_weight_kernel():
+    """One-launch fwd+dgrad weight pack vs the eager permute/flip chains."""
+    ext = _backend.hip_extension()
+    torch.manual_seed(4)
+    w = torch.randn(24, 50, 3, 3, device="cuda").bfloat16()
+    fwd = torch.empty(24, 9 * 50, dtype=torch.bfloat16, device="cuda")
+    dgr = torch.empty(50, 9 * 24, dtype=torch.bfloat16, device="cuda")
+    ext.pack_conv_weight(w, fwd, dgr)
+    ref_f = w.permute(0, 2, 3, 1).reshape(24, -1)
+    ref_d = torch.flip(w, dims=(2, 3)).permute(1, 2, 3, 0).reshape(50, -1)
+    assert torch.equal(fwd, ref_f.contiguous())
+    assert torch.equal(dgr, ref_d.contiguous())
+
+
 def test_tr16_probe_delivery_map():
     """Pin the ds_read_b64_tr_b16 semantics the wgrad kernel is built on:
     with per-lane address base + l*8B over a [64][16]-short LDS image, lane l
