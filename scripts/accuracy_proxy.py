@@ -89,7 +89,7 @@ def main():
     ap.add_argument("--steps", type=int, default=600)
     ap.add_argument("--batch", type=int, default=8)
     ap.add_argument("--nstack", type=int, default=2)
-    ap.add_argument("--lr", type=float, default=7e-4)
+    ap.add_argument("--lr", type=float, default=4e-4)
     ap.add_argument("--eval", type=int, default=16)
     ap.add_argument("--max-people", type=int, default=2)
     ap.add_argument("--seed", type=int, default=11)
@@ -120,6 +120,7 @@ def main():
           f"batch {args.batch}, lr {args.lr}, {args.steps} steps, "
           f"rendered synthetic scenes (<= {args.max_people} people)", flush=True)
     t0 = time.time()
+    skipped = 0
     for step in range(args.steps):
         imgs, mms, hms = [], [], []
         for b in range(args.batch):
@@ -129,16 +130,28 @@ def main():
             hms.append(torch.from_numpy(np.ascontiguousarray(hm)))
         batch = tuple(torch.stack(t).to(device=device, dtype=dtype)
                       for t in (imgs, mms, hms))
-        # warmup then flat
-        lr = args.lr * min((step + 1) / 50.0, 1.0)
+        # warmup, then halve every 500 steps (flat 7e-4 diverged at ~step 60
+        # in one kernel-rounding configuration — bf16 + no loss scaling needs
+        # headroom)
+        lr = args.lr * min((step + 1) / 50.0, 1.0) * (0.5 ** (step // 500))
         for g in optimizer.param_groups:
             g["lr"] = lr
         optimizer.zero_grad(set_to_none=True)
         loss = model(batch)
+        lf = float(loss.detach())
+        # reference-style abnormal-batch guard (train_distributed.py:259-261):
+        # a non-finite / exploded loss skips backward+step so weights stay
+        # finite
+        if not (lf == lf and abs(lf) < 1e6):
+            skipped += 1
+            print(f"step {step:5d}  loss {lf} — SKIPPED (guard)", flush=True)
+            if skipped > args.steps // 10:
+                raise SystemExit("too many skipped batches — diverged")
+            continue
         loss.backward()
         optimizer.step()
         if step % 25 == 0 or step == args.steps - 1:
-            print(f"step {step:5d}  loss {float(loss):9.3f}  lr {lr:.2e}  "
+            print(f"step {step:5d}  loss {lf:9.3f}  lr {lr:.2e}  "
                   f"{time.time() - t0:6.1f}s", flush=True)
 
     # evaluation through the full pipeline on HELD-OUT scenes
