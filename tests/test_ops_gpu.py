@@ -266,6 +266,8 @@ def test_mfma_conv_wgrad_vs_reference():
         (2, 3, 64, 64, 7, 2, 1),      # 7x7 s2 stem (elementwise gather)
         (3, 192, 320, 20, 3, 1, 1),   # odd M tail
         (2, 16, 64, 32, 4, 1, 1),     # Cin=16 multi-tap (s2d stem shape)
+        (4, 256, 128, 128, 1, 1, 1),  # wide (64,128) tile (M >= 65536, 1x1)
+        (4, 50, 128, 128, 1, 1, 1),   # wide tile + cin tail (ELEM)
     ]
     for n, cin, cout, hw, k, s, d in cases:
         torch.manual_seed(3)
