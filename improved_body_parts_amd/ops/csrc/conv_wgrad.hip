@@ -59,6 +59,7 @@ struct WgradParams {
   int kd_tiles, co_tiles, chunks;
   long long chunk_len;        // multiple of 64
   int step_dho, step_dwo;     // 64 / Wo, 64 % Wo (m-walk carry steps)
+  int kd_cb;                  // Cin/TKD when Cin % TKD == 0 (tile reorder), else 0
 };
 
 __device__ __forceinline__ int wg_imgrow(int m) {
@@ -92,8 +93,14 @@ __launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
   // tile coords: blockIdx = ((chunk * kd_tiles) + kdt) * co_tiles + cot
   int bid = blockIdx.x;
   const int cot = bid % p.co_tiles; bid /= p.co_tiles;
-  const int kdt = bid % p.kd_tiles;
+  int kdt = bid % p.kd_tiles;
   const int chunk = bid / p.kd_tiles;
+  if (p.kd_cb > 0) {
+    // launch the taps of one cin-block adjacently: the 9(taps) x co_tiles
+    // blocks re-reading the same x window then run concurrently and share L2
+    const int ntap = p.kd_tiles / p.kd_cb;
+    kdt = (kdt % ntap) * p.kd_cb + kdt / ntap;
+  }
   const int kd0 = kdt * TKD;
   const int co0 = cot * TCO;
   const long long m_begin = (long long)chunk * p.chunk_len;
@@ -421,6 +428,7 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   p.KD = (int)(KH * KW * Cin);
   p.step_dho = 64 / p.Wo;
   p.step_dwo = 64 % p.Wo;
+  p.kd_cb = 0;  // set below once TKD is chosen
   // Tile selection. The wide-spatial 3x3 layers are L3-BANDWIDTH-bound on
   // re-reads: with a TCOxTKD grid each x byte is read (taps * Cout/TCO)
   // times and each dy byte (KD/TKD) times per chunk (~9.7 GB for the
@@ -439,6 +447,9 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   const int TCO = big ? 128 : (wide ? 128 : 64);
   p.kd_tiles = (p.KD + TKD - 1) / TKD;
   p.co_tiles = (int)((Cout + TCO - 1) / TCO);
+  if (KH * KW > 1 && Cin % TKD == 0) p.kd_cb = (int)(Cin / TKD);
+  if (const char* e = getenv("IBP_WGRAD_NOREORDER"))
+    if (e[0] == '1') p.kd_cb = 0;
   // split M into chunks for parallelism: aim ~768 blocks, chunk length a
   // multiple of 64; every chunk slice is fully written (plain stores) before
   // the combine reduces them in fixed order -> deterministic
