@@ -70,7 +70,10 @@ __device__ __forceinline__ int lds_off(int row, int k) {
   return row * BK + (k ^ ((row & 7) << 3));
 }
 
-template <int BM, int BN>
+// DEEP: two-stage register prefetch — wins ONLY on short-K shapes (the 1x1
+// layers at <= 6 K-chunks never reach pipeline steady state); on long-K 3x3
+// shapes the extra register bank measured 10-15% slower.
+template <int BM, int BN, bool DEEP = false>
 __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   __shared__ unsigned short lds_a[2][BM * BK];
   __shared__ unsigned short lds_b[2][BN * BK];
@@ -136,12 +139,12 @@ __global__ __launch_bounds__(256, 2) void conv_mfma_kernel(ConvParams p) {
   const int nk = min(nk_chunk, nk_total - k_begin);
   if (nk <= 0) return;
 
-  // single-stage register prefetch: a 2-deep variant (as in conv_wgrad.hip)
-  // was measured 10-15% SLOWER here — the fwd gather stages 2-4x more
-  // register bytes per thread than wgrad, and the doubled banks cost more in
-  // scheduling/pressure than the extra latency cover buys
   unsigned short a_reg0[A_ELEMS];
   unsigned short b_reg0[B_ELEMS];
+  // second bank referenced only under `if constexpr (DEEP)` — fully
+  // eliminated in the default instantiations
+  unsigned short a_reg1[A_ELEMS];
+  unsigned short b_reg1[B_ELEMS];
 
   // ---- incremental tap-walking state for the A gather ----------------------
   // k = (kh*KW + kw)*Cin + ci decodes with ONE division pair at setup; each
@@ -311,16 +314,35 @@ a_done:
   };
 
   // ---- main loop: register-staged double buffer ----------------------------
-  load_chunk(k_begin, a_reg0, b_reg0);
-  write_chunk(0, a_reg0, b_reg0);
-  __syncthreads();
-  for (int t = 0; t < nk; ++t) {
-    if (t + 1 < nk) load_chunk(k_begin + t + 1, a_reg0, b_reg0);
-    compute(t & 1);
-    // writing buf[(t+1)&1] is safe without a barrier: its last readers were
-    // separated by the end-of-iteration barrier of step t-1
-    if (t + 1 < nk) write_chunk((t + 1) & 1, a_reg0, b_reg0);
+  if constexpr (DEEP) {
+    // 2-deep prefetch: chunk t+2's loads issue at iteration t, land at t+1
+    load_chunk(k_begin, a_reg0, b_reg0);
+    write_chunk(0, a_reg0, b_reg0);
+    if (nk > 1) load_chunk(k_begin + 1, a_reg1, b_reg1);
     __syncthreads();
+    for (int t = 0; t < nk; ++t) {
+      if (t + 2 < nk) load_chunk(k_begin + t + 2, a_reg0, b_reg0);
+      compute(0);
+      if (t + 1 < nk) write_chunk(1, a_reg1, b_reg1);
+      __syncthreads();
+      if (++t >= nk) break;
+      if (t + 2 < nk) load_chunk(k_begin + t + 2, a_reg1, b_reg1);
+      compute(1);
+      if (t + 1 < nk) write_chunk(0, a_reg0, b_reg0);
+      __syncthreads();
+    }
+  } else {
+    load_chunk(k_begin, a_reg0, b_reg0);
+    write_chunk(0, a_reg0, b_reg0);
+    __syncthreads();
+    for (int t = 0; t < nk; ++t) {
+      if (t + 1 < nk) load_chunk(k_begin + t + 1, a_reg0, b_reg0);
+      compute(t & 1);
+      // writing buf[(t+1)&1] is safe without a barrier: its last readers
+      // were separated by the end-of-iteration barrier of step t-1
+      if (t + 1 < nk) write_chunk((t + 1) & 1, a_reg0, b_reg0);
+      __syncthreads();
+    }
   }
 
   // ---- epilogue: C/D lane map col = lane&15, row = (lane>>4)*4 + r ---------
@@ -511,14 +533,30 @@ Tensor conv_mfma_fwd(const Tensor& x, const Tensor& w_packed, int64_t N,
     p.ws = nullptr;
   }
   dim3 grid(ntiles * splitk), block(256);
+  // short-K blocks (the 1x1 layers: <= 6 chunks) never reach pipeline
+  // steady state with single-stage prefetch — use the 2-deep variant there
+  const int nk_block = (nk_total + splitk - 1) / splitk;
+  const bool deep = nk_block <= 6;
   if (BM == 128 && BN == 128) {
-    hipLaunchKernelGGL((ibp::conv_mfma_kernel<128, 128>), grid, block, 0, stream, p);
+    if (deep)
+      hipLaunchKernelGGL((ibp::conv_mfma_kernel<128, 128, true>), grid, block, 0, stream, p);
+    else
+      hipLaunchKernelGGL((ibp::conv_mfma_kernel<128, 128>), grid, block, 0, stream, p);
   } else if (BM == 128) {
-    hipLaunchKernelGGL((ibp::conv_mfma_kernel<128, 64>), grid, block, 0, stream, p);
+    if (deep)
+      hipLaunchKernelGGL((ibp::conv_mfma_kernel<128, 64, true>), grid, block, 0, stream, p);
+    else
+      hipLaunchKernelGGL((ibp::conv_mfma_kernel<128, 64>), grid, block, 0, stream, p);
   } else if (BM == 64) {
-    hipLaunchKernelGGL((ibp::conv_mfma_kernel<64, 64>), grid, block, 0, stream, p);
+    if (deep)
+      hipLaunchKernelGGL((ibp::conv_mfma_kernel<64, 64, true>), grid, block, 0, stream, p);
+    else
+      hipLaunchKernelGGL((ibp::conv_mfma_kernel<64, 64>), grid, block, 0, stream, p);
   } else {
-    hipLaunchKernelGGL((ibp::conv_mfma_kernel<32, 64>), grid, block, 0, stream, p);
+    if (deep)
+      hipLaunchKernelGGL((ibp::conv_mfma_kernel<32, 64, true>), grid, block, 0, stream, p);
+    else
+      hipLaunchKernelGGL((ibp::conv_mfma_kernel<32, 64>), grid, block, 0, stream, p);
   }
   if (splitk > 1) {
     long long total = (long long)p.M * Cout;
