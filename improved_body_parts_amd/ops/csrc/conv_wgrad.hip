@@ -44,13 +44,14 @@ typedef short short8 __attribute__((ext_vector_type(8)));
 typedef float floatx4 __attribute__((ext_vector_type(4)));
 typedef unsigned short ushortv8 __attribute__((ext_vector_type(8)));
 
-// LDS geometry: (TKD+TCO)/16 subtiles, each [64 rows][16 ch] plus an 8-B pad.
-// Stride 2056 B -> inter-subtile bank shift of 2 dwords: an 8-lane b128 write
-// group (4 subtiles x 2 pixels, banks q*2 + perm(m)*8 mod 32) hits 8 DISTINCT
-// banks (the previous 16-B pad gave shift 4 -> 2-way conflicts, PMC
-// SQ_LDS_BANK_CONFLICT 6.0e8). tr reads are unaffected: each read's 256
-// contiguous bytes span 64 distinct banks from any 8-B-aligned base.
-constexpr int SUB_SHORTS = 64 * 16 + 4;   // 1028 shorts = 2056 B (8-B aligned)
+// LDS geometry: (TKD+TCO)/16 subtiles, each [64 rows][16 ch] plus a 16-B pad
+// so consecutive subtiles start on different write-bank groups. The stride
+// MUST stay 16-B aligned: an 8-B pad (tried for a conflict-free write
+// stagger) misaligns every b128 staging write on odd subtiles — measured
+// 0.31-0.46x, far worse than the 2-way conflicts it removed. A fully
+// conflict-free stagger is impossible here: {q*s} mod 32 is arithmetic, the
+// needed residue set is not.
+constexpr int SUB_SHORTS = 64 * 16 + 8;   // 1032 shorts = 2064 B (16-B aligned)
 
 struct WgradParams {
   const unsigned short* x;    // NHWC bf16
