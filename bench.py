@@ -239,6 +239,16 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t)
 
+    if os.environ.get("IBP_DDP_TIMING") == "1" and args.mode == "train" \
+            and use_cuda and distributed:
+        # untimed extra steps measuring exposed vs total all-reduce time
+        # (stderr only — stdout stays the single JSON contract line)
+        step(0)
+        step(1)
+        reducer_t = reducer.last_timing if reducer is not None else None
+        if rank == 0 and reducer_t:
+            print(f"[ddp-timing] {reducer_t}", file=sys.stderr, flush=True)
+
     images = batch * args.steps * world_size
     value = images / elapsed
     ms_per_step = elapsed / args.steps * 1000.0

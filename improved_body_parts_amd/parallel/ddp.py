@@ -99,6 +99,12 @@ class GradReducer:
         self._hooks = [p.register_post_accumulate_grad_hook(self._on_grad)
                        for p in params]
         self.require_sync = True
+        # IBP_DDP_TIMING=1: measure exposed (non-overlapped) all-reduce time
+        # with CUDA events — evidence for the backward/communication overlap
+        import os
+        self.timing = (os.environ.get("IBP_DDP_TIMING") == "1"
+                       and device.type == "cuda")
+        self.last_timing = None
         self._reset_pending()
 
     # ------------------------------------------------------------------ hooks
@@ -106,6 +112,7 @@ class GradReducer:
         for b in self.buckets:
             b.pending = len(b.params)
             b.work = None
+            b.launch_ev = None
 
     def _on_grad(self, param):
         if not self.require_sync or self.world_size <= 1:
@@ -113,6 +120,9 @@ class GradReducer:
         b = self._param_bucket[id(param)]
         b.pending -= 1
         if b.pending == 0:
+            if self.timing:
+                b.launch_ev = torch.cuda.Event(enable_timing=True)
+                b.launch_ev.record()
             b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
                                      group=self.group, async_op=True)
 
@@ -127,6 +137,10 @@ class GradReducer:
         if self.world_size <= 1 or not self.require_sync:
             return
         inv = 1.0 / self.world_size
+        t_back = None
+        if self.timing:
+            t_back = torch.cuda.Event(enable_timing=True)
+            t_back.record()  # ~end of backward compute
         for b in self.buckets:
             if b.work is None and b.pending > 0:
                 # a parameter did not receive a gradient this step (e.g. an
@@ -137,6 +151,24 @@ class GradReducer:
             if b.work is not None:
                 b.work.wait()
             b.flat.mul_(inv)
+        if self.timing:
+            t_done = torch.cuda.Event(enable_timing=True)
+            t_done.record()
+            torch.cuda.synchronize()
+            first = None
+            for b in self.buckets:
+                if b.launch_ev is not None:
+                    first = b.launch_ev
+                    break
+            self.last_timing = {
+                # collectives outstanding after backward finished = EXPOSED
+                "exposed_ms": t_back.elapsed_time(t_done),
+                # first bucket launch -> all reduced = total comm span
+                # (span >> exposed means the overlap is doing its job)
+                "comm_span_ms": (first.elapsed_time(t_done)
+                                 if first is not None else 0.0),
+                "buckets": len(self.buckets),
+            }
 
     def flat_grads(self):
         """The flat bucket buffers (for the fused multi-tensor optimizer)."""
