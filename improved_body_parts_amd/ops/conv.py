@@ -55,6 +55,27 @@ def _conv_wgrad(x, dy, w_shape, stride, padding, dilation):
     return dw
 
 
+_BIAS_FOLD = {}
+
+
+def _bias_fold(weight, bias):
+    """(ones, bias.float()) for the bias-only epilogue, cached against the
+    bias version — a fresh pair per call was ~60 FillFunctor/copy launches
+    per inference step (the 1x1 heads carry biases)."""
+    import weakref
+    key = id(bias)
+    entry = _BIAS_FOLD.get(key)
+    ver = bias._version
+    if entry is not None and entry[0] == ver:
+        return entry[1], entry[2]
+    scale = torch.ones(weight.shape[0], device=bias.device, dtype=torch.float32)
+    shift = bias.detach().float()
+    if entry is None:
+        weakref.finalize(bias, _BIAS_FOLD.pop, key, None)
+    _BIAS_FOLD[key] = (ver, scale, shift)
+    return scale, shift
+
+
 def _tick_running_stats(bn_mod):
     """The finalize kernel updates running stats in place without going
     through ATen — tick their version counters so the eval-path folded-BN
@@ -94,11 +115,7 @@ class ConvBnActFn(torch.autograd.Function):
                     shift = beta.float() - bn_mod.running_mean.float() * scale
                     bn_mod._ibp_folded = (ver, scale, shift)
             elif bias is not None:
-                scale = torch.ones(weight.shape[0], device=x.device,
-                                   dtype=torch.float32)
-                shift = bias.float()
-            else:
-                scale = shift = None
+                scale, shift = _bias_fold(weight, bias)
             if scale is not None or act or residual is not None \
                     or residual_post is not None:
                 y = conv_kernels.conv_fwd(x, weight, stride, padding, dilation,

@@ -38,7 +38,8 @@ Tensor se_reduce(const Tensor& a, const c10::optional<Tensor>& b, int64_t N,
 Tensor se_scale(const Tensor& x, const Tensor& s, const c10::optional<Tensor>& addc,
                 int64_t N, int64_t HW, int64_t C);
 Tensor se_gate(const Tensor& pooled, const Tensor& w1, const Tensor& b1,
-               const Tensor& w2, const Tensor& b2, double slope);
+               const Tensor& w2, const Tensor& b2, double slope,
+               double pool_scale);
 
 // loss.hip
 Tensor focal_l2_fwd(const Tensor& pred, const Tensor& gt, const Tensor& mask,
@@ -103,7 +104,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("upsample2x_bwd", &upsample2x_bwd);
   m.def("se_reduce", &se_reduce);
   m.def("se_scale", &se_scale);
-  m.def("se_gate", &se_gate, "fused GAP-gate: sigmoid(W2@leaky(W1@p+b1)+b2)");
+  m.def("se_gate", &se_gate, "fused GAP-gate: sigmoid(W2@leaky(W1@p+b1)+b2)",
+        py::arg("pooled"), py::arg("w1"), py::arg("b1"), py::arg("w2"),
+        py::arg("b2"), py::arg("slope"), py::arg("pool_scale") = 1.0);
   m.def("focal_l2_fwd", &focal_l2_fwd);
   m.def("focal_l2_bwd", &focal_l2_bwd);
   m.def("conv_mfma_fwd", &conv_mfma_fwd, "MFMA implicit-GEMM conv forward",

@@ -133,11 +133,13 @@ __global__ void se_gate_kernel(const float* __restrict__ pooled,
                                const T* __restrict__ w1, const T* __restrict__ b1,
                                const T* __restrict__ w2, const T* __restrict__ b2,
                                float* __restrict__ s, int C, int CH,
-                               float slope) {
+                               float slope, float pool_scale) {
   __shared__ float pl[768 + 64];
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
-  for (int c = tid; c < C; c += blockDim.x) pl[c] = pooled[(long long)n * C + c];
+  // pool_scale folds the GAP division (sums -> mean) into this kernel
+  for (int c = tid; c < C; c += blockDim.x)
+    pl[c] = pooled[(long long)n * C + c] * pool_scale;
   __syncthreads();
   if (tid < CH) {
     float acc = ldf(b1 + tid);
@@ -259,7 +261,8 @@ Tensor se_reduce(const Tensor& a, const c10::optional<Tensor>& b, int64_t N,
 }
 
 Tensor se_gate(const Tensor& pooled, const Tensor& w1, const Tensor& b1,
-               const Tensor& w2, const Tensor& b2, double slope) {
+               const Tensor& w2, const Tensor& b2, double slope,
+               double pool_scale) {
   TORCH_CHECK(pooled.is_cuda() && pooled.dim() == 2 &&
               pooled.scalar_type() == at::ScalarType::Float);
   int64_t N = pooled.size(0), C = pooled.size(1);
@@ -288,7 +291,8 @@ Tensor se_gate(const Tensor& pooled, const Tensor& w1, const Tensor& b1,
                        reinterpret_cast<const T*>(b1.data_ptr()),
                        reinterpret_cast<const T*>(w2.data_ptr()),
                        reinterpret_cast<const T*>(b2.data_ptr()),
-                       s.data_ptr<float>(), (int)C, (int)CH, (float)slope);
+                       s.data_ptr<float>(), (int)C, (int)CH, (float)slope,
+                       (float)pool_scale);
   });
   return s;
 }

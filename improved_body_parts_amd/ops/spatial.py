@@ -116,9 +116,10 @@ def se_layer_hip(x, fc1, fc2):
         ext = hip_extension()
         xc = x.contiguous(memory_format=_CL)
         n, c, h_, w_ = xc.shape
-        pooled = ext.se_reduce(xc, None, n, h_ * w_, c) / float(h_ * w_)
+        pooled = ext.se_reduce(xc, None, n, h_ * w_, c)
+        # the GAP division rides the gate kernel (pool_scale) — one less launch
         s = ext.se_gate(pooled, fc1.weight, fc1.bias, fc2.weight, fc2.bias,
-                        LEAKY_SLOPE)
+                        LEAKY_SLOPE, 1.0 / float(h_ * w_))
         return ext.se_scale(xc, s, None, n, h_ * w_, c)
     pooled = _SeGapFn.apply(x)                       # [N, C] fp32
     h = F.leaky_relu(fc1(pooled.to(fc1.weight.dtype)), LEAKY_SLOPE)
