@@ -459,11 +459,12 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   // multiple of 64; every chunk slice is fully written (plain stores) before
   // the combine reduces them in fixed order -> deterministic
   const int tiles = p.kd_tiles * p.co_tiles;
-  long long max_chunks = (p.M + 63) / 64;
-  // r14 sweep on (256ch, 3x3, 128^2): 768 -> 0.758 ms, 3072 -> 0.712,
-  // 6144 -> 0.691 — shorter chunks win (load balance + more blocks in
-  // flight); the combine is a cheap recursive tree so chunk count is nearly
-  // free. Small layers cap at M/64 chunks anyway.
+  // r14 sweep on (256ch, 3x3, 128^2): 768 blocks -> 0.758 ms, 3072 -> 0.712,
+  // 6144 -> 0.691 — more chunks win there (load balance). But a chunk must
+  // still run >= ~32 pipeline stages or the block is pure prologue (r15: the
+  // unconditional 6144 target dropped 1-stage chunks to 0.28-0.48x) — cap
+  // chunks so every block keeps >= 32 x 64-m stages.
+  long long max_chunks = std::max<long long>(p.M / (64 * 32), 1);
   int target_blocks = 6144;
   if (const char* e = getenv("IBP_WGRAD_BLOCKS")) target_blocks = atoi(e);
   long long want = (target_blocks + tiles - 1) / tiles;
