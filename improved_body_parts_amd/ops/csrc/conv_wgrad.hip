@@ -80,15 +80,23 @@ __device__ __forceinline__ short4v tr16_read(const unsigned short* p) {
 
 // ELEM compiles in the tap-crossing per-element path (stem, Cin % 16 != 0,
 // partial last kd-tile) — its precomputed pe_pack walk costs ~16 VGPRs, so
-// the clean-tiled hot layers instantiate ELEM=false and keep occupancy 4
-template <int TKD, int TCO, bool ELEM>
+// the clean-tiled hot layers instantiate ELEM=false and keep occupancy 4.
+// NWAVES=8 (512 threads) is the 128x128 "mid" tile: same waves/SIMD as the
+// 64x64 tile but 25% fewer LDS tr-read bytes per FLOP (the 64-tile at occ 4
+// demands ~4.5x the LDS read rate its MFMA shadow covers — wgrad's binding
+// resource on the wide 3x3 layers).
+template <int TKD, int TCO, bool ELEM, int NWAVES = 4>
 __global__
-__launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
+__launch_bounds__(NWAVES * 64,
+                  (TKD == 64 && TCO == 64) ? 3
+                                           : (NWAVES == 8 ? 4 : 2))
+    void conv_wgrad_kernel(
     WgradParams p) {
   constexpr int ASUB = TKD / 16;
   constexpr int BSUB = TCO / 16;
   constexpr int NSUB = ASUB + BSUB;
-  constexpr int NPIECE = NSUB / 4;        // 16-ch pieces per thread per stage
+  constexpr int NPIECE = NSUB / NWAVES;   // 16-ch pieces per thread per stage
+  constexpr int WCOLS = NWAVES / 2;       // waves as 2 x WCOLS
   __shared__ unsigned short lds[2][NSUB * SUB_SHORTS];
 
   const int tid = threadIdx.x;
@@ -112,11 +120,11 @@ __launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
   const long long m_end = min(m_begin + p.chunk_len, p.M);
   if (m_begin >= m_end) return;
 
-  // wave sub-tile: 2x2 waves of (TKD/2)x(TCO/2)
-  constexpr int AFRAG = TKD / 32;
-  constexpr int BFRAG = TCO / 32;
-  const int wm = wave >> 1;       // dW-row half
-  const int wn = wave & 1;        // cout half
+  // wave sub-tile: 2 x WCOLS waves of (TKD/2) x (TCO/WCOLS)
+  constexpr int AFRAG = (TKD / 2) / 16;
+  constexpr int BFRAG = (TCO / WCOLS) / 16;
+  const int wm = wave / WCOLS;    // dW-row half
+  const int wn = wave % WCOLS;    // cout column group
 
   floatx4 acc[AFRAG][BFRAG];
   #pragma unroll
@@ -124,10 +132,10 @@ __launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
     #pragma unroll
     for (int j = 0; j < BFRAG; ++j) acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
 
-  // ---- staging geometry: thread t handles pixel m0+(t>>2) and NPIECE
-  // subtiles q, q+4, ... (q = t&3); first ASUB subtiles are A, rest B
-  const int st_m = tid >> 2;
-  const int st_q = tid & 3;
+  // ---- staging geometry: thread t handles pixel m0 + t/NWAVES and NPIECE
+  // subtiles q, q+NWAVES, ... (q = t%NWAVES); first ASUB subtiles are A
+  const int st_m = tid / NWAVES;
+  const int st_q = tid % NWAVES;
   const int st_row = wg_imgrow(st_m);
 
   // per-piece dW-row bases (tap + cin) — divisions happen ONCE here; the
@@ -149,7 +157,7 @@ __launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
   }
   #pragma unroll
   for (int pi = 0; pi < NPIECE; ++pi) {
-    const int sub = st_q + 4 * pi;
+    const int sub = st_q + NWAVES * pi;
     if (sub < ASUB) {
       const int k0 = kd0 + 16 * sub;
       const int f = p.Cin > 0 ? k0 / p.Cin : 0;
@@ -279,7 +287,7 @@ __launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
   auto write_stage = [&](int buf, unsigned short (&regs)[NPIECE][16]) {
     #pragma unroll
     for (int pi = 0; pi < NPIECE; ++pi) {
-      const int sub = st_q + 4 * pi;
+      const int sub = st_q + NWAVES * pi;
       unsigned short* dst = &lds[buf][sub * SUB_SHORTS + st_row * 16];
       #pragma unroll
       for (int v = 0; v < 2; ++v)
@@ -322,6 +330,19 @@ __launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
   // stage t+2's loads are issued at iteration t, written to LDS at t+1 —
   // each load has two full compute phases to land
   const long long n_stages = (m_end - m_begin + 63) >> 6;
+  if constexpr (NWAVES == 8) {
+    // single-bank prefetch: the 8-wave block hides latency by wave count;
+    // a second register bank pushed the 128-VGPR budget into spills
+    load_stage(regs0);
+    write_stage(0, regs0);
+    __syncthreads();
+    for (long long t = 0; t < n_stages; ++t) {
+      if (t + 1 < n_stages) load_stage(regs0);
+      compute((int)(t & 1));
+      if (t + 1 < n_stages) write_stage((int)((t + 1) & 1), regs0);
+      __syncthreads();
+    }
+  } else {
   load_stage(regs0);                           // stage 0
   write_stage(0, regs0);
   if (n_stages > 1) load_stage(regs1);         // stage 1
@@ -337,6 +358,7 @@ __launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
     if (t + 1 < n_stages) write_stage(0, regs0);
     __syncthreads();
   }
+  }
 
   // ---- epilogue: D lane map col = lane&15, row = (lane>>4)*4 + r
   const int ecol = lane & 15;
@@ -346,7 +368,7 @@ __launch_bounds__(256, (TKD == 64 && TCO == 64) ? 3 : 2) void conv_wgrad_kernel(
   for (int i = 0; i < AFRAG; ++i) {
     #pragma unroll
     for (int j = 0; j < BFRAG; ++j) {
-      const int co = co0 + wn * (TCO / 2) + j * 16 + ecol;
+      const int co = co0 + wn * (TCO / WCOLS) + j * 16 + ecol;
       if (co >= p.Cout) continue;
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -448,8 +470,12 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   // the tap factor, not Cout/TCO, dominates traffic) -> 1x1 only
   const bool wide = !big && KH == 1 && KW == 1 && Cout >= 128 &&
                     p.M >= 65536;
-  const int TKD = big ? 128 : 64;
-  const int TCO = big ? 128 : (wide ? 128 : 64);
+  // mid tile (128x128 @ 8 waves): for the LDS-read-bound wide-spatial KxK
+  // layers — same waves/SIMD as 64x64 but 25% fewer tr-read bytes per FLOP
+  const bool mid = !big && !wide && KH * KW > 1 && p.KD >= 1024 &&
+                   Cout >= 128 && p.M >= 65536;
+  const int TKD = (big || mid) ? 128 : 64;
+  const int TCO = (big || mid || wide) ? 128 : 64;
   p.kd_tiles = (p.KD + TKD - 1) / TKD;
   p.co_tiles = (int)((Cout + TCO - 1) / TCO);
   if (KH * KW > 1 && Cin % TKD == 0) p.kd_cb = (int)(Cin / TKD);
@@ -482,7 +508,13 @@ Tensor conv_mfma_wgrad(const Tensor& x, const Tensor& dy, int64_t N, int64_t H,
   // tap-crossing pieces exist iff Cin is not 16-aligned or the last kd-tile
   // is partial — only then compile in the per-element path (costs VGPRs)
   const bool elem = (Cin % 16 != 0) || (p.KD % TKD != 0);
-  if (big && elem) {
+  if (mid && elem) {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128, true, 8>), grid,
+                       dim3(512), 0, stream, p);
+  } else if (mid) {
+    hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128, false, 8>), grid,
+                       dim3(512), 0, stream, p);
+  } else if (big && elem) {
     hipLaunchKernelGGL((ibp::conv_wgrad_kernel<128, 128, true>), grid, block,
                        0, stream, p);
   } else if (big) {
