@@ -21,9 +21,10 @@
 //   * row permutation when staging pixel m (within a 32-row block):
 //     imgrow = 16*((m>>2)&1) + 4*(m>>3) + (m&3), so that the two tr reads of a
 //     fragment (base, base+512B) deliver exactly m-slices g*8+0..3 / g*8+4..7.
-//   * block tile TKDxTCO: (128,128) for the big layers (32 MFMA : 32 tr reads
-//     per wave-stage — MFMA-bound) and (64,64) for heads/merges/small KD
-//     (8 MFMA : 16 tr reads), 4 waves as 2x2; double-buffered 64-m stages.
+//   * block tile TKDxTCO x NWAVES: (64,64)x4 for heads/merges/small layers
+//     (occ 4), (64,128)x4 for wide 1x1s, and the (128,128)x8 "mid" tile for
+//     the LDS-read-bound wide-spatial KxK layers (same waves/SIMD as 64x64,
+//     25% fewer tr-read bytes per FLOP); double-buffered 64-m stages.
 //   * the per-stage pixel decomposition m -> (n, ho, wo) advances by +64 with
 //     carry steps (no 64-bit divisions in the loop).
 //   * split-M determinism: each m-chunk writes its own fp32 workspace slice
