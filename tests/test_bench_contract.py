@@ -74,3 +74,22 @@ def test_bench_torchrun_two_ranks_cpu():
     assert j["n_gpus"] == 2
     assert j["config"]["parallelism"] == "dp2"
     assert j["config"]["global_batch"] == 2  # batch 1 per rank, aggregated
+
+
+@pytest.mark.timeout(900)
+def test_bench_torchrun_four_ranks_cpu():
+    """The driver's N=4 launch shape (SCALE run) over gloo."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29527", BENCH, "--gpus", "4", "--steps", "1",
+         "--warmup", "0", "--nstack", "1", "--input", "128", "--batch", "1",
+         "--mode", "train"],
+        capture_output=True, text=True, timeout=840, cwd=REPO, env=env)
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
+    j = _last_json_line(r.stdout)
+    assert j["n_gpus"] == 4
+    assert j["config"]["global_batch"] == 4

@@ -2,6 +2,7 @@
 
 Run on an MI355X: python -m pytest tests -m gpu -x -q
 """
+import numpy as np
 import pytest
 import torch
 import torch.nn.functional as F
@@ -218,6 +219,41 @@ def test_conv_post_act_residuals(training):
         dy_ref = (2 * y.detach().float())
         _assert_rel(r1g.grad, dy_ref, 2e-2, "dres_post identity")
         _assert_rel(r2g.grad, dy_ref, 2e-2, "dres_post2 identity")
+
+
+def test_mfma_conv_fuzz_shapes():
+    """Randomized shapes across the kernel envelope (tile heuristics, tap
+    walk, tails) vs the library conv — guards the gather's many branches."""
+    from improved_body_parts_amd.ops import conv_kernels
+    rng = np.random.RandomState(20)
+    for trial in range(12):
+        k = int(rng.choice([1, 1, 3, 3, 5, 7]))
+        cin = int(rng.choice([8, 16, 24, 48, 64, 96, 128, 50] if k == 1
+                             else [8, 16, 48, 64, 128]))
+        cout = int(rng.choice([16, 50, 64, 77, 128, 192]))
+        hw = int(rng.choice([7, 8, 12, 16, 24, 32, 48]))
+        n = int(rng.randint(1, 4))
+        d = int(rng.choice([1, 1, 1, 2, 3])) if k == 3 else 1
+        s = int(rng.choice([1, 1, 2]))
+        pad = (k - 1) // 2 * d
+        if hw + 2 * pad < d * (k - 1) + 1:
+            continue
+        torch.manual_seed(trial)
+        x = torch.randn(n, cin, hw, hw, device="cuda").bfloat16() \
+            .contiguous(memory_format=CL)
+        w = (torch.randn(cout, cin, k, k, device="cuda") * 0.05).bfloat16()
+        y = conv_kernels.conv_fwd(x, w, (s, s), (pad, pad), (d, d))
+        assert y is not None, (n, cin, cout, hw, k, s, d)
+        ref = F.conv2d(x.float(), w.float(), None, s, pad, d)
+        _assert_rel(y, ref, 1.5e-2, f"fuzz fwd {(n, cin, cout, hw, k, s, d)}")
+        if s == 1:
+            dy = (torch.randn_like(ref) * 0.1).bfloat16() \
+                .contiguous(memory_format=CL)
+            dw = conv_kernels.conv_wgrad(x, dy, w.shape, (s, s), (pad, pad),
+                                         (d, d))
+            refw = torch.nn.grad.conv2d_weight(x.float(), w.shape, dy.float(),
+                                               s, pad, d)
+            _assert_rel(dw, refw, 2e-2, f"fuzz wgrad {(n, cin, cout, hw, k, d)}")
 
 
 def test_pack_conv_weight_kernel():
