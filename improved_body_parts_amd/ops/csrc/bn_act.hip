@@ -355,28 +355,6 @@ __global__ void bn_act_bwd_reduce_bf16v8(
   }
 }
 
-// coefficient precompute for the apply kernel: the training BN backward
-//   dx = gamma*invstd*(dpre - sum_dpre/M - xhat*sum_dxhat/M)
-// folds to dx = P[c]*dpre + Q[c]*x + R[c]; one thread per channel.
-__global__ void bn_bwd_coeffs_kernel(
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    const float* __restrict__ gamma, const float* __restrict__ sum_dpre,
-    const float* __restrict__ sum_dxhat, float* __restrict__ pqr, int C,
-    float invM) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float is = invstd[c];
-  float A = gamma[c] * is;
-  float P = A, Q = 0.f, R = 0.f;
-  if (sum_dpre) {
-    Q = -A * is * sum_dxhat[c] * invM;
-    R = -A * sum_dpre[c] * invM - Q * mean[c];
-  }
-  pqr[c] = P;
-  pqr[C + c] = Q;
-  pqr[2 * C + c] = R;
-}
-
 typedef float float4v __attribute__((ext_vector_type(4)));
 
 // vectorized bf16 BN backward apply: dx = P[c]*dpre + Q[c]*x + R[c].
