@@ -171,11 +171,153 @@ __global__ void se_scale_kernel(const T* __restrict__ x, const float* __restrict
   }
 }
 
+// ---- vectorized bf16 variants (C % 8 == 0): 16-B accesses, one pixel
+// decomposition per 8 channels — the scalar forms were per-element
+// div/mod + 2-B-access bound (~4x off bandwidth in the r2 profile) --------
+typedef unsigned short ushort8v __attribute__((ext_vector_type(8)));
+
+__global__ void maxpool2x2_fwd_bf16v8(const ushort8v* __restrict__ x,
+                                      ushort8v* __restrict__ y,
+                                      unsigned char* __restrict__ arg,
+                                      int N, int H, int W, int C8) {
+  const int Ho = H >> 1, Wo = W >> 1;
+  const long long total = (long long)N * Ho * Wo * C8;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int c8 = (int)(i % C8);
+    long long p = i / C8;
+    const int wo = (int)(p % Wo);
+    long long q = p / Wo;
+    const int ho = (int)(q % Ho);
+    const int n = (int)(q / Ho);
+    const ushort8v* base =
+        x + (((long long)n * H + 2 * ho) * W + 2 * wo) * C8 + c8;
+    ushort8v v00 = base[0];
+    ushort8v v01 = base[C8];
+    ushort8v v10 = base[(long long)W * C8];
+    ushort8v v11 = base[(long long)W * C8 + C8];
+    ushort8v out;
+    unsigned char* ab = arg + i * 8;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float m = us2f(v00[j]);
+      int a = 0;
+      float f01 = us2f(v01[j]), f10 = us2f(v10[j]), f11 = us2f(v11[j]);
+      if (f01 > m) { m = f01; a = 1; }
+      if (f10 > m) { m = f10; a = 2; }
+      if (f11 > m) { m = f11; a = 3; }
+      out[j] = f2us(m);
+      ab[j] = (unsigned char)a;
+    }
+    y[i] = out;
+  }
+}
+
+__global__ void maxpool2x2_bwd_bf16v8(const ushort8v* __restrict__ dy,
+                                      const unsigned char* __restrict__ arg,
+                                      ushort8v* __restrict__ dx,
+                                      int N, int H, int W, int C8) {
+  const int Ho = H >> 1, Wo = W >> 1;
+  const long long total = (long long)N * H * W * C8;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int c8 = (int)(i % C8);
+    long long p = i / C8;
+    const int w = (int)(p % W);
+    long long q = p / W;
+    const int h = (int)(q % H);
+    const int n = (int)(q / H);
+    ushort8v out = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (h < 2 * Ho && w < 2 * Wo) {
+      const long long o =
+          (((long long)n * Ho + (h >> 1)) * Wo + (w >> 1)) * C8 + c8;
+      const int pos = ((h & 1) << 1) | (w & 1);
+      const ushort8v g = dy[o];
+      const unsigned char* ab = arg + o * 8;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (ab[j] == pos) out[j] = g[j];
+    }
+    dx[i] = out;
+  }
+}
+
+__global__ void upsample2x_fwd_bf16v8(const ushort8v* __restrict__ x,
+                                      ushort8v* __restrict__ y,
+                                      int N, int H, int W, int C8) {
+  const int Ho = H * 2, Wo = W * 2;
+  const long long total = (long long)N * Ho * Wo * C8;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int c8 = (int)(i % C8);
+    long long p = i / C8;
+    const int wo = (int)(p % Wo);
+    long long q = p / Wo;
+    const int ho = (int)(q % Ho);
+    const int n = (int)(q / Ho);
+    y[i] = x[(((long long)n * H + (ho >> 1)) * W + (wo >> 1)) * C8 + c8];
+  }
+}
+
+__global__ void upsample2x_bwd_bf16v8(const ushort8v* __restrict__ dy,
+                                      ushort8v* __restrict__ dx,
+                                      int N, int H, int W, int C8) {
+  const int Ho = H * 2, Wo = W * 2;
+  const long long total = (long long)N * H * W * C8;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int c8 = (int)(i % C8);
+    long long p = i / C8;
+    const int w = (int)(p % W);
+    long long q = p / W;
+    const int h = (int)(q % H);
+    const int n = (int)(q / H);
+    const ushort8v* base =
+        dy + (((long long)n * Ho + 2 * h) * Wo + 2 * w) * C8 + c8;
+    ushort8v d0 = base[0], d1 = base[C8];
+    ushort8v d2 = base[(long long)Wo * C8], d3 = base[(long long)Wo * C8 + C8];
+    ushort8v out;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      out[j] = f2us((us2f(d0[j]) + us2f(d1[j])) +
+                    (us2f(d2[j]) + us2f(d3[j])));
+    dx[i] = out;
+  }
+}
+
+__global__ void se_scale_bf16v8(const ushort8v* __restrict__ x,
+                                const float* __restrict__ s,
+                                const float* __restrict__ addc,
+                                ushort8v* __restrict__ y,
+                                int N, long long HW, int C8) {
+  const long long total = (long long)N * HW * C8;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int c8 = (int)(i % C8);
+    const int n = (int)(i / (HW * C8));
+    const float* sc = s + (long long)n * C8 * 8 + c8 * 8;
+    const float* ad = addc ? addc + (long long)n * C8 * 8 + c8 * 8 : nullptr;
+    ushort8v xv = x[i];
+    ushort8v out;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float v = us2f(xv[j]) * sc[j];
+      if (ad) v += ad[j];
+      out[j] = f2us(v);
+    }
+    y[i] = out;
+  }
+}
+
 }  // namespace ibp
 
 using torch::Tensor;
 static inline hipStream_t cur_stream2() {
   return at::hip::getCurrentHIPStream().stream();
+}
+
+static inline bool v8_ok(const Tensor& t, int64_t C) {
+  return t.scalar_type() == at::ScalarType::BFloat16 && C % 8 == 0;
 }
 
 std::vector<Tensor> maxpool2x2_fwd(const Tensor& x, int64_t N, int64_t H, int64_t W,
@@ -184,6 +326,15 @@ std::vector<Tensor> maxpool2x2_fwd(const Tensor& x, int64_t N, int64_t H, int64_
   Tensor arg = torch::empty({N, H / 2, W / 2, C},
                             x.options().dtype(torch::kUInt8));
   dim3 block(256), grid(ibp::grid_1d(y.numel(), 256, 8192));
+  if (v8_ok(x, C)) {
+    dim3 g8(ibp::grid_1d(y.numel() / 8, 256, 8192));
+    hipLaunchKernelGGL(ibp::maxpool2x2_fwd_bf16v8, g8, block, 0, cur_stream2(),
+                       reinterpret_cast<const ibp::ushort8v*>(x.data_ptr()),
+                       reinterpret_cast<ibp::ushort8v*>(y.data_ptr()),
+                       arg.data_ptr<unsigned char>(), (int)N, (int)H, (int)W,
+                       (int)(C / 8));
+    return {y, arg};
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
       x.scalar_type(), "maxpool2x2_fwd", [&] {
     hipLaunchKernelGGL(ibp::maxpool2x2_fwd_kernel<scalar_t>, grid, block, 0,
@@ -198,6 +349,15 @@ Tensor maxpool2x2_bwd(const Tensor& dy, const Tensor& arg, int64_t N, int64_t H,
                       int64_t W, int64_t C) {
   Tensor dx = torch::empty({N, H, W, C}, dy.options());
   dim3 block(256), grid(ibp::grid_1d(dx.numel(), 256, 8192));
+  if (v8_ok(dy, C)) {
+    dim3 g8(ibp::grid_1d(dx.numel() / 8, 256, 8192));
+    hipLaunchKernelGGL(ibp::maxpool2x2_bwd_bf16v8, g8, block, 0, cur_stream2(),
+                       reinterpret_cast<const ibp::ushort8v*>(dy.data_ptr()),
+                       arg.data_ptr<unsigned char>(),
+                       reinterpret_cast<ibp::ushort8v*>(dx.data_ptr()),
+                       (int)N, (int)H, (int)W, (int)(C / 8));
+    return dx;
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
       dy.scalar_type(), "maxpool2x2_bwd", [&] {
     hipLaunchKernelGGL(ibp::maxpool2x2_bwd_kernel<scalar_t>, grid, block, 0,
@@ -212,6 +372,14 @@ Tensor maxpool2x2_bwd(const Tensor& dy, const Tensor& arg, int64_t N, int64_t H,
 Tensor upsample2x_fwd(const Tensor& x, int64_t N, int64_t H, int64_t W, int64_t C) {
   Tensor y = torch::empty({N, H * 2, W * 2, C}, x.options());
   dim3 block(256), grid(ibp::grid_1d(y.numel(), 256, 8192));
+  if (v8_ok(x, C)) {
+    dim3 g8(ibp::grid_1d(y.numel() / 8, 256, 8192));
+    hipLaunchKernelGGL(ibp::upsample2x_fwd_bf16v8, g8, block, 0, cur_stream2(),
+                       reinterpret_cast<const ibp::ushort8v*>(x.data_ptr()),
+                       reinterpret_cast<ibp::ushort8v*>(y.data_ptr()),
+                       (int)N, (int)H, (int)W, (int)(C / 8));
+    return y;
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
       x.scalar_type(), "upsample2x_fwd", [&] {
     hipLaunchKernelGGL(ibp::upsample2x_fwd_kernel<scalar_t>, grid, block, 0,
@@ -225,6 +393,14 @@ Tensor upsample2x_fwd(const Tensor& x, int64_t N, int64_t H, int64_t W, int64_t 
 Tensor upsample2x_bwd(const Tensor& dy, int64_t N, int64_t H, int64_t W, int64_t C) {
   Tensor dx = torch::empty({N, H, W, C}, dy.options());
   dim3 block(256), grid(ibp::grid_1d(dx.numel(), 256, 8192));
+  if (v8_ok(dy, C)) {
+    dim3 g8(ibp::grid_1d(dx.numel() / 8, 256, 8192));
+    hipLaunchKernelGGL(ibp::upsample2x_bwd_bf16v8, g8, block, 0, cur_stream2(),
+                       reinterpret_cast<const ibp::ushort8v*>(dy.data_ptr()),
+                       reinterpret_cast<ibp::ushort8v*>(dx.data_ptr()),
+                       (int)N, (int)H, (int)W, (int)(C / 8));
+    return dx;
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
       dy.scalar_type(), "upsample2x_bwd", [&] {
     hipLaunchKernelGGL(ibp::upsample2x_bwd_kernel<scalar_t>, grid, block, 0,
@@ -301,6 +477,16 @@ Tensor se_scale(const Tensor& x, const Tensor& s, const c10::optional<Tensor>& a
                 int64_t N, int64_t HW, int64_t C) {
   Tensor y = torch::empty_like(x);
   dim3 block(256), grid(ibp::grid_1d(x.numel(), 256, 8192));
+  if (v8_ok(x, C)) {
+    dim3 g8(ibp::grid_1d(x.numel() / 8, 256, 8192));
+    hipLaunchKernelGGL(ibp::se_scale_bf16v8, g8, block, 0, cur_stream2(),
+                       reinterpret_cast<const ibp::ushort8v*>(x.data_ptr()),
+                       s.data_ptr<float>(),
+                       addc.has_value() ? addc->data_ptr<float>() : nullptr,
+                       reinterpret_cast<ibp::ushort8v*>(y.data_ptr()),
+                       (int)N, HW, (int)(C / 8));
+    return y;
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
       x.scalar_type(), "se_scale", [&] {
     using T = scalar_t;
