@@ -106,9 +106,12 @@ __global__ void upsample2x_bwd_kernel(const T* __restrict__ dy, T* __restrict__ 
 
 // ---- squeeze-excitation helpers -------------------------------------------
 // channel reduction r[n][c] = sum_hw a[n][h][w][c] * (b ? b[n][h][w][c] : 1)
+// Partials go to a [N][rows][C] workspace with plain stores; a column-sum
+// stage reduces rows (the previous atomicAdd form serialised rows-way per
+// (n,c) address and needed a zero-init launch).
 template <typename T, bool PROD>
 __global__ void se_reduce_kernel(const T* __restrict__ a, const T* __restrict__ b,
-                                 float* __restrict__ out, int N, long long HW, int C) {
+                                 float* __restrict__ ws, int N, long long HW, int C) {
   int c = blockIdx.y * blockDim.x + threadIdx.x;
   if (c >= C) return;
   int n = blockIdx.z;
@@ -120,7 +123,33 @@ __global__ void se_reduce_kernel(const T* __restrict__ a, const T* __restrict__ 
     if (PROD) v *= ldf(pb + m * C + c);
     s += v;
   }
-  atomicAdd(&out[(long long)n * C + c], s);
+  ws[((long long)n * gridDim.x + blockIdx.x) * C + c] = s;
+}
+
+// out[p][c] = sum_r ws[p][r][c] (same shape as bn_act.hip's colsum; kept
+// local — device code cannot cross TUs under -fno-gpu-rdc)
+__global__ __launch_bounds__(1024) void se_colsum_kernel(
+    const float* __restrict__ ws, float* __restrict__ out, int rows, int C,
+    int planes) {
+  __shared__ float red[1024];
+  const int total = C * planes;
+  const int cl = threadIdx.x & 63;
+  const int rl = threadIdx.x >> 6;
+  const int i = blockIdx.x * 64 + cl;
+  float a = 0.f;
+  if (i < total) {
+    const int p = i / C;
+    const int c = i - p * C;
+    const float* src = ws + (long long)p * rows * C + c;
+    for (int r = rl; r < rows; r += 16) a += src[(long long)r * C];
+  }
+  red[threadIdx.x] = a;
+  __syncthreads();
+  if (rl == 0 && i < total) {
+    #pragma unroll
+    for (int g = 1; g < 16; ++g) a += red[g * 64 + cl];
+    out[i] = a;
+  }
 }
 
 // y = x * s[n][c] (+ optional add[n][c] broadcast)
@@ -415,10 +444,12 @@ Tensor upsample2x_bwd(const Tensor& dy, int64_t N, int64_t H, int64_t W, int64_t
 // SE backward ds (b = the other operand)
 Tensor se_reduce(const Tensor& a, const c10::optional<Tensor>& b, int64_t N,
                  int64_t HW, int64_t C) {
-  Tensor out = torch::zeros({N, C}, a.options().dtype(torch::kFloat32));
+  Tensor out = torch::empty({N, C}, a.options().dtype(torch::kFloat32));
   dim3 block(256);
   int rows = (int)std::min<long long>((HW + 63) / 64, 512);
   dim3 grid(rows, (C + 255) / 256, N);
+  Tensor ws = torch::empty({N * (int64_t)rows * C},
+                           a.options().dtype(torch::kFloat32));
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
       a.scalar_type(), "se_reduce", [&] {
     using T = scalar_t;
@@ -426,13 +457,17 @@ Tensor se_reduce(const Tensor& a, const c10::optional<Tensor>& b, int64_t N,
       hipLaunchKernelGGL((ibp::se_reduce_kernel<T, true>), grid, block, 0,
                          cur_stream2(), reinterpret_cast<const T*>(a.data_ptr()),
                          reinterpret_cast<const T*>(b->data_ptr()),
-                         out.data_ptr<float>(), (int)N, HW, (int)C);
+                         ws.data_ptr<float>(), (int)N, HW, (int)C);
     } else {
       hipLaunchKernelGGL((ibp::se_reduce_kernel<T, false>), grid, block, 0,
                          cur_stream2(), reinterpret_cast<const T*>(a.data_ptr()),
-                         nullptr, out.data_ptr<float>(), (int)N, HW, (int)C);
+                         nullptr, ws.data_ptr<float>(), (int)N, HW, (int)C);
     }
   });
+  dim3 cgrid(((int)(N * C) + 63) / 64);
+  hipLaunchKernelGGL(ibp::se_colsum_kernel, cgrid, dim3(1024), 0, cur_stream2(),
+                     ws.data_ptr<float>(), out.data_ptr<float>(), rows, (int)C,
+                     (int)N);
   return out;
 }
 
