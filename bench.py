@@ -272,7 +272,9 @@ def main():
             "config": {
                 "model": f"{args.nstack}-stage IMHN @{args.input}x{args.input}",
                 "mode": args.mode,
-                "train_data": args.data if args.mode == "train" else None,
+                # device-gt needs a GPU; report what actually ran
+                "train_data": (args.data if use_cuda else "prestaged")
+                if args.mode == "train" else None,
                 "global_batch": batch * world_size,
                 "input": args.input,
                 "parallelism": f"dp{world_size}",
