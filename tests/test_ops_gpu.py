@@ -328,6 +328,8 @@ def test_mfma_conv_wgrad_vs_reference():
         (2, 16, 64, 32, 4, 1, 1),     # Cin=16 multi-tap (s2d stem shape)
         (4, 256, 128, 128, 1, 1, 1),  # wide (64,128) tile (M >= 65536, 1x1)
         (4, 50, 128, 128, 1, 1, 1),   # wide tile + cin tail (ELEM)
+        (4, 128, 128, 128, 3, 1, 1),  # mid (128,128)x8-wave tile (KxK, M>=65536)
+        (4, 128, 128, 128, 3, 1, 5),  # mid tile, dilation 5
     ]
     for n, cin, cout, hw, k, s, d in cases:
         torch.manual_seed(3)
