@@ -58,7 +58,11 @@ class GaussianSmoothing(nn.Module):
             kernel = kernel * (1 / (std * math.sqrt(2 * math.pi))
                                * torch.exp(-(((mgrid - mean) / std) ** 2) / 2))
         kernel = kernel / kernel.sum()
-        kernel = kernel.view(1, 1, *kernel.shape).repeat(channels, *([1] * (kernel.dim())))
+        # NOTE: view first, THEN take dim() — evaluating kernel.dim() inside
+        # the same statement reads the pre-view tensor (latent crash for any
+        # dim, caught by tests/test_utils.py)
+        kernel = kernel.view(1, 1, *kernel.shape)
+        kernel = kernel.repeat(channels, *([1] * (kernel.dim() - 1)))
         self.register_buffer("weight", kernel)
         self.groups = channels
         self.conv = {1: F.conv1d, 2: F.conv2d, 3: F.conv3d}[dim]
