@@ -256,6 +256,37 @@ def test_mfma_conv_fuzz_shapes():
             _assert_rel(dw, refw, 2e-2, f"fuzz wgrad {(n, cin, cout, hw, k, d)}")
 
 
+def test_mfma_conv_rectangular_inputs():
+    """Non-square H x W — the real inference path pads to arbitrary x64
+    aspect ratios (reference predict() / padRightDownCorner)."""
+    from improved_body_parts_amd.ops import conv_kernels
+    for (n, cin, cout, h, w, k, s, d) in [
+        (2, 64, 128, 24, 40, 3, 1, 1),
+        (1, 128, 128, 16, 48, 3, 1, 3),
+        (2, 256, 64, 12, 20, 1, 1, 1),
+        (1, 3, 64, 96, 160, 7, 2, 1),   # stem s2d, rectangular
+    ]:
+        torch.manual_seed(8)
+        x = torch.randn(n, cin, h, w, device="cuda").bfloat16() \
+            .contiguous(memory_format=CL)
+        wt = (torch.randn(cout, cin, k, k, device="cuda") * 0.05).bfloat16()
+        pad = (k - 1) // 2 * d
+        y = conv_kernels.conv_fwd(x, wt, (s, s), (pad, pad), (d, d))
+        ref = F.conv2d(x.float(), wt.float(), None, s, pad, d)
+        _assert_rel(y, ref, 1.5e-2, f"rect fwd {(n, cin, cout, h, w, k, s, d)}")
+        dy = (torch.randn_like(ref) * 0.1).bfloat16().contiguous(memory_format=CL)
+        dw = conv_kernels.conv_wgrad(x, dy, wt.shape, (s, s), (pad, pad), (d, d))
+        refw = torch.nn.grad.conv2d_weight(x.float(), wt.shape, dy.float(),
+                                           s, pad, d)
+        _assert_rel(dw, refw, 2e-2, f"rect wgrad {(n, cin, cout, h, w, k, s, d)}")
+        if s == 1:
+            dx = conv_kernels.conv_dgrad(dy, wt, x.shape, (s, s), (pad, pad),
+                                         (d, d))
+            refx = torch.nn.grad.conv2d_input(x.shape, wt.float(), dy.float(),
+                                              s, pad, d)
+            _assert_rel(dx, refx, 1.5e-2, f"rect dgrad {(n, cin, cout, h, w, k)}")
+
+
 def test_conv_kernels_bitwise_deterministic():
     """The split-K (fwd) and split-M (wgrad) designs claim determinism: no
     atomics, fixed-order slice reduction. Verify bit equality across runs."""
