@@ -134,3 +134,25 @@ def test_lr_schedule():
     lr_decay = adjust_learning_rate(o, epoch=15, iters_done=0, iters_per_epoch=100,
                                     base_lr=1.0, warmup_epochs=3, decay_every=15)
     assert lr_decay == pytest.approx(0.2)
+
+
+def test_trainer_resume_continues_epoch(tiny_setup):
+    """save -> resume in a fresh Trainer -> epochs continue where they left
+    off with weights and optimizer state restored (reference
+    train_distributed.py:149-197)."""
+    cfg, opt, ds, tmp = tiny_setup
+    tr = Trainer(opt, cfg, ds, rank=0, world_size=1, num_workers=0,
+                 checkpoint_dir=str(tmp), device=torch.device("cpu"))
+    tr.fit(1)
+    w0 = tr.model.posenet.pre.conv1.weight.detach().clone()
+
+    tr2 = Trainer(opt, cfg, ds, rank=0, world_size=1, num_workers=0,
+                  checkpoint_dir=str(tmp), device=torch.device("cpu"))
+    tr2.resume(str(tmp / "PoseNet_0_epoch.pth"))
+    assert tr2.start_epoch == 1
+    assert torch.equal(tr2.model.posenet.pre.conv1.weight.detach(), w0)
+    # optimizer momentum state came back too
+    states = [s for s in tr2.optimizer.state.values() if s]
+    assert states, "optimizer state not restored"
+    tr2.fit(2)  # runs epoch 1 only
+    assert (tmp / "PoseNet_1_epoch.pth").exists()
