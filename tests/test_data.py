@@ -208,3 +208,113 @@ def test_coco_h5_iterator_augmented_runs():
     img, mask_miss, labels = ds[0]
     assert img.shape == (cfg.height, cfg.width, 3)
     assert torch.isfinite(labels).all()
+
+
+# ---------------------------------------------------------------------------
+# offline h5 BUILDER exercised with faked pycocotools / h5py (round-1 verdict:
+# "written, never run") — main-person selection, mask_miss/mask_all, layout
+# ---------------------------------------------------------------------------
+def test_build_coco_h5_with_fakes(tmp_path, monkeypatch):
+    import json
+    import sys
+    import types
+
+    from PIL import Image
+
+    # --- tiny COCO-format annotation world --------------------------------
+    W = H = 64
+    img_dir = tmp_path / "imgs"
+    img_dir.mkdir()
+    Image.fromarray(np.zeros((H, W, 3), np.uint8)).save(img_dir / "a.jpg")
+
+    anns = [
+        # main person: enough keypoints + area
+        {"id": 1, "image_id": 10, "num_keypoints": 10, "area": 40 * 40,
+         "bbox": [4, 4, 40, 40], "iscrowd": 0,
+         "segmentation": [[4, 4, 44, 4, 44, 44, 4, 44]],
+         "keypoints": sum([[8 + j, 10 + j, 2] for j in range(17)], [])},
+        # small person: masked out of mask_miss, not a main person
+        {"id": 2, "image_id": 10, "num_keypoints": 3, "area": 8 * 8,
+         "bbox": [50, 50, 8, 8], "iscrowd": 0,
+         "segmentation": [[50, 50, 58, 50, 58, 58, 50, 58]],
+         "keypoints": sum([[52, 52, 2]] * 17, [])},
+    ]
+
+    class FakeCOCO:
+        def __init__(self, ann_file):
+            self.imgs = {10: {"id": 10, "file_name": "a.jpg",
+                              "height": H, "width": W}}
+
+        def getAnnIds(self, imgIds):
+            return [a["id"] for a in anns if a["image_id"] == imgIds]
+
+        def loadAnns(self, ids):
+            return [a for a in anns if a["id"] in ids]
+
+        def loadImgs(self, img_id):
+            return [self.imgs[img_id]]
+
+    def fr_py_objects(seg, h, w):
+        return seg
+
+    def decode(rle):
+        xs = rle[0][0::2]
+        ys = rle[0][1::2]
+        m = np.zeros((H, W), np.uint8)
+        m[min(ys):max(ys), min(xs):max(xs)] = 1
+        return m
+
+    fake_coco_mod = types.ModuleType("pycocotools.coco")
+    fake_coco_mod.COCO = FakeCOCO
+    fake_mask_mod = types.ModuleType("pycocotools.mask")
+    fake_mask_mod.frPyObjects = fr_py_objects
+    fake_mask_mod.decode = decode
+    fake_root = types.ModuleType("pycocotools")
+    fake_root.coco = fake_coco_mod
+    fake_root.mask = fake_mask_mod
+    monkeypatch.setitem(sys.modules, "pycocotools", fake_root)
+    monkeypatch.setitem(sys.modules, "pycocotools.coco", fake_coco_mod)
+    monkeypatch.setitem(sys.modules, "pycocotools.mask", fake_mask_mod)
+
+    # --- in-memory h5 recorder --------------------------------------------
+    store = {}
+
+    class FakeGroup(dict):
+        def create_dataset(self, name, data=None, **kw):
+            self[name] = np.asarray(data) if not isinstance(data, str) else data
+
+    class FakeFile:
+        def __init__(self, path, mode):
+            self.groups = store
+
+        def create_group(self, name):
+            g = FakeGroup()
+            self.groups[name] = g
+            return g
+
+        def __enter__(self):
+            return self
+
+        def __exit__(self, *a):
+            return False
+
+    import improved_body_parts_amd.data.coco as coco_mod
+    monkeypatch.setattr(coco_mod, "h5py",
+                        types.SimpleNamespace(File=FakeFile))
+
+    out = coco_mod.build_coco_h5("unused.json", str(img_dir),
+                                 str(tmp_path / "out.h5"), image_size=64)
+    assert out.endswith("out.h5")
+    assert set(store) == {"dataset", "images", "masks"}
+    # exactly ONE main person selected (the small one fails the filters)
+    assert list(store["dataset"].keys()) == ["0"]
+    meta = json.loads(store["dataset"]["0"])
+    assert meta["image"] == "a.jpg"
+    assert len(meta["joints"]) == 2          # but ALL persons' joints kept
+    # visibility recode: coco v=2 -> 1 (visible)
+    assert meta["joints"][0][0][2] == 1.0
+    mask_miss, mask_all = store["masks"]["a.jpg"]
+    # the small unusable person is zeroed in mask_miss, present in mask_all
+    assert mask_miss[54, 54] == 0.0 and mask_miss[20, 20] == 1.0
+    assert mask_all[54, 54] == 1.0 and mask_all[20, 20] == 1.0
+    assert store["images"]["a.jpg"].shape == (H, W, 3)
