@@ -35,6 +35,10 @@ def main():
     ap.add_argument("--config", default="Canonical")
     ap.add_argument("-n", type=int, default=4)
     ap.add_argument("--save-dir", default=None)
+    ap.add_argument("--bench", action="store_true",
+                    help="augmentation/GT-generation throughput (reference "
+                         "data/mydataset.py test_augmentation_speed; the "
+                         "reference reports ~40 samples/s per CPU process)")
     args = ap.parse_args()
 
     from improved_body_parts_amd.config import COCOSourceConfig, GetConfig
@@ -49,6 +53,18 @@ def main():
         ds = MyDataset(config, COCOSourceConfig(args.h5), augment=False)
         get = lambda i: ds[i]
         print(f"{len(ds)} records in {args.h5}")
+
+    if args.bench:
+        import time
+        get(0)  # warm caches
+        t0 = time.perf_counter()
+        n = max(args.n, 16)
+        for i in range(n):
+            get(i)
+        dt = time.perf_counter() - t0
+        print(f"{n / dt:.1f} samples/s single process "
+              f"(transform + GT generation; reference: ~40/s)")
+        return
 
     for i in range(args.n):
         img, mask_miss, labels = (t.numpy() for t in get(i))
